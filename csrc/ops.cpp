@@ -1,0 +1,190 @@
+// Torch bindings for the tosem2021_amd gfx950 HIP kernels.
+//
+// Host-only translation unit: device code lives in the *.hip files, exposed
+// through C-linkage launchers.  Uses the HIP-native ATen stream API directly
+// (c10::hip) — no CUDA names, no hipify translation of this file is needed.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+hipError_t ln_fwd_launch(const void*, const void*, const void*, void*, void*,
+                         void*, int, int, float, int, hipStream_t);
+hipError_t ln_bwd_launch(const void*, const void*, const void*, const void*,
+                         const void*, void*, void*, void*, int, int, int,
+                         hipStream_t);
+hipError_t colsum_launch(const void*, void*, int, int, hipStream_t);
+hipError_t bias_gelu_fwd_launch(const void*, const void*, void*, long, int,
+                                int, hipStream_t);
+hipError_t bias_gelu_bwd_launch(const void*, const void*, const void*, void*,
+                                void*, long, int, int, hipStream_t);
+hipError_t softmax_fwd_launch(const void*, const void*, void*, long, int, int,
+                              float, int, hipStream_t);
+hipError_t softmax_bwd_launch(const void*, const void*, void*, long, int,
+                              float, int, hipStream_t);
+hipError_t adamw_launch(void*, const void*, int, void*, void*, void*, long,
+                        float, float, float, float, float, int, float, int,
+                        hipStream_t);
+}
+
+namespace {
+
+#define CHECK_HIP(call)                                                   \
+  do {                                                                    \
+    hipError_t e_ = (call);                                               \
+    TORCH_CHECK(e_ == hipSuccess, "HIP error: ", hipGetErrorString(e_));  \
+  } while (0)
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.device().is_cuda(), name, " must be on GPU");
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be f32");
+  TORCH_CHECK(t.device().is_cuda(), name, " must be on GPU");
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta, double eps) {
+  check_bf16(x, "x"); check_bf16(gamma, "gamma"); check_bf16(beta, "beta");
+  const int D = (int)x.size(-1);
+  const long N = x.numel() / D;
+  TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8, got ", D);
+  auto y = torch::empty_like(x);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({N}, opts);
+  auto rstd = torch::empty({N}, opts);
+  int grid = (int)std::min<long>((N + 3) / 4, 1024);
+  CHECK_HIP(ln_fwd_launch(x.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
+                          y.data_ptr(), mean.data_ptr(), rstd.data_ptr(),
+                          (int)N, D, (float)eps, grid, cur_stream()));
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor gamma, torch::Tensor mean,
+                                         torch::Tensor rstd) {
+  check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(gamma, "gamma");
+  const int D = (int)x.size(-1);
+  const long N = x.numel() / D;
+  auto dx = torch::empty_like(x);
+  int grid = (int)std::min<long>((N + 3) / 4, 1024);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto ws_dg = torch::empty({grid, D}, opts);
+  auto ws_db = torch::empty({grid, D}, opts);
+  CHECK_HIP(ln_bwd_launch(dy.data_ptr(), x.data_ptr(), gamma.data_ptr(),
+                          mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
+                          ws_dg.data_ptr(), ws_db.data_ptr(), (int)N, D, grid,
+                          cur_stream()));
+  auto dgamma = torch::empty({D}, opts);
+  auto dbeta = torch::empty({D}, opts);
+  CHECK_HIP(colsum_launch(ws_dg.data_ptr(), dgamma.data_ptr(), grid, D,
+                          cur_stream()));
+  CHECK_HIP(colsum_launch(ws_db.data_ptr(), dbeta.data_ptr(), grid, D,
+                          cur_stream()));
+  return {dx, dgamma, dbeta};
+}
+
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b) {
+  check_bf16(x, "x"); check_bf16(b, "b");
+  const int D = (int)x.size(-1);
+  const long n = x.numel();
+  TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
+  auto y = torch::empty_like(x);
+  int grid = (int)std::min<long>((n / 8 + 255) / 256, 2048);
+  CHECK_HIP(bias_gelu_fwd_launch(x.data_ptr(), b.data_ptr(), y.data_ptr(), n,
+                                 D, grid, cur_stream()));
+  return y;
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor b) {
+  check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(b, "b");
+  const int D = (int)x.size(-1);
+  const long n = x.numel();
+  auto dx = torch::empty_like(x);
+  int grid = (int)std::min<long>((n / 8 + 255) / 256, 1024);
+  auto ws = torch::empty({grid, D}, x.options().dtype(torch::kFloat32));
+  CHECK_HIP(bias_gelu_bwd_launch(dy.data_ptr(), x.data_ptr(), b.data_ptr(),
+                                 dx.data_ptr(), ws.data_ptr(), n, D, grid,
+                                 cur_stream()));
+  auto dbias = torch::empty({D}, x.options().dtype(torch::kFloat32));
+  CHECK_HIP(colsum_launch(ws.data_ptr(), dbias.data_ptr(), grid, D,
+                          cur_stream()));
+  return {dx, dbias};
+}
+
+torch::Tensor softmax_fwd(torch::Tensor scores, c10::optional<torch::Tensor> mask,
+                          double scale) {
+  check_bf16(scores, "scores");
+  const int Lk = (int)scores.size(-1);
+  const long n_rows = scores.numel() / Lk;
+  TORCH_CHECK(Lk % 8 == 0, "Lk must be a multiple of 8");
+  int H_Lq = 1;
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    TORCH_CHECK(scores.dim() == 4, "masked softmax expects [B,H,Lq,Lk]");
+    TORCH_CHECK(mask->size(0) == scores.size(0) && mask->size(-1) == Lk,
+                "mask must be [B,Lk]");
+    H_Lq = (int)(scores.size(1) * scores.size(2));
+    mptr = mask->data_ptr();
+  }
+  auto p = torch::empty_like(scores);
+  int grid = (int)std::min<long>((n_rows + 3) / 4, 2048);
+  CHECK_HIP(softmax_fwd_launch(scores.data_ptr(), mptr, p.data_ptr(), n_rows,
+                               Lk, H_Lq, (float)scale, grid, cur_stream()));
+  return p;
+}
+
+torch::Tensor softmax_bwd(torch::Tensor dp, torch::Tensor p, double scale) {
+  check_bf16(dp, "dp"); check_bf16(p, "p");
+  const int Lk = (int)p.size(-1);
+  const long n_rows = p.numel() / Lk;
+  auto ds = torch::empty_like(p);
+  int grid = (int)std::min<long>((n_rows + 3) / 4, 2048);
+  CHECK_HIP(softmax_bwd_launch(dp.data_ptr(), p.data_ptr(), ds.data_ptr(),
+                               n_rows, Lk, (float)scale, grid, cur_stream()));
+  return ds;
+}
+
+void adamw_step(torch::Tensor p, torch::Tensor grad, torch::Tensor m,
+                torch::Tensor v, torch::Tensor master, double lr, double beta1,
+                double beta2, double eps, double wd, long step,
+                double grad_scale) {
+  check_bf16(p, "p"); check_f32(m, "m"); check_f32(v, "v");
+  check_f32(master, "master");
+  TORCH_CHECK(grad.is_contiguous() && grad.device().is_cuda(), "bad grad");
+  int gf32 = grad.scalar_type() == torch::kFloat32;
+  TORCH_CHECK(gf32 || grad.scalar_type() == torch::kBFloat16,
+              "grad must be f32 or bf16");
+  const long n = p.numel();
+  TORCH_CHECK(n % 4 == 0, "flat parameter buffer must be padded to 4 elems");
+  TORCH_CHECK(grad.numel() == n && m.numel() == n && v.numel() == n &&
+              master.numel() == n, "size mismatch");
+  int grid = (int)std::min<long>((n / 4 + 255) / 256, 2048);
+  CHECK_HIP(adamw_launch(p.data_ptr(), grad.data_ptr(), gf32, m.data_ptr(),
+                         v.data_ptr(), master.data_ptr(), n, (float)lr,
+                         (float)beta1, (float)beta2, (float)eps, (float)wd,
+                         (int)step, (float)grad_scale, grid, cur_stream()));
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (bf16, gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd");
+  m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU fwd");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU bwd");
+  m.def("softmax_fwd", &softmax_fwd, "fused scaled masked softmax fwd");
+  m.def("softmax_bwd", &softmax_bwd, "fused softmax bwd");
+  m.def("adamw_step", &adamw_step, "fused AdamW over flat params");
+}

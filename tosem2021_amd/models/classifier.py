@@ -1,0 +1,206 @@
+"""MLTC — the learned test-case classifier (the framework's flagship model).
+
+A bidirectional transformer encoder over tokenized test code / assertion text
+that predicts the reference study's taxonomy labels (RQs/taxonomy_test2.csv:1):
+
+  * 19 test strategies   (RQ1_tests.csv rows)      — multi-label
+  * 21 quality properties (tests_prop_rq3.csv:1)   — multi-label
+  * 9 ML workflow stages  (RQ1_tests.csv columns)  — single-label
+  * 4 test methods        (tests_methods.csv:2-5)  — single-label
+
+MI355X-first design: all parameters bf16 (flat-packed by the trainer), GEMMs
+through hipBLASLt (torch.matmul / F.linear on ROCm), and the non-GEMM hot ops
+(LayerNorm, bias+GeLU, masked softmax) are hand-written gfx950 HIP kernels
+(csrc/*.hip) behind tosem2021_amd.ops.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from tosem2021_amd import ops
+
+# Label spaces (mirrors tosem2021_amd.extract.schema; duplicated sizes only)
+N_STRATEGIES = 19
+N_PROPERTIES = 21
+N_STAGES = 9
+N_METHODS = 4
+
+HEAD_SIZES = {
+    "strategy": N_STRATEGIES,   # multi-label (BCE)
+    "property": N_PROPERTIES,   # multi-label (BCE)
+    "stage": N_STAGES,          # single-label (CE)
+    "method": N_METHODS,        # single-label (CE)
+}
+MULTILABEL_HEADS = ("strategy", "property")
+
+
+@dataclass
+class MLTCConfig:
+    vocab_size: int = 32768
+    d_model: int = 1024
+    n_heads: int = 16
+    n_layers: int = 12
+    d_ff: int = 4096
+    max_seq: int = 512
+    dropout: float = 0.0
+    layer_norm_eps: float = 1e-5
+    heads: Dict[str, int] = field(default_factory=lambda: dict(HEAD_SIZES))
+
+    @property
+    def head_dim(self) -> int:
+        assert self.d_model % self.n_heads == 0
+        return self.d_model // self.n_heads
+
+
+CONFIGS: Dict[str, MLTCConfig] = {
+    # flagship: the bench.py / BASELINE config
+    "mltc-base": MLTCConfig(),
+    # small config for fast CPU tests / smoke
+    "mltc-tiny": MLTCConfig(vocab_size=512, d_model=128, n_heads=4, n_layers=2,
+                            d_ff=256, max_seq=64),
+}
+
+
+class FusedLayerNorm(nn.Module):
+    def __init__(self, d: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(d))
+        self.bias = nn.Parameter(torch.zeros(d))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.fused_layernorm(x, self.weight, self.bias, self.eps)
+
+
+class Attention(nn.Module):
+    """Multi-head bidirectional self-attention.
+
+    QK^T and PV run as batched hipBLASLt GEMMs (MFMA); the scaled masked
+    softmax between them is the fused gfx950 kernel (csrc/softmax.hip).
+    """
+
+    def __init__(self, cfg: MLTCConfig):
+        super().__init__()
+        self.n_heads = cfg.n_heads
+        self.head_dim = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self.qkv = nn.Linear(cfg.d_model, 3 * cfg.d_model, bias=True)
+        self.proj = nn.Linear(cfg.d_model, cfg.d_model, bias=True)
+
+    def forward(self, x, mask_bias: Optional[torch.Tensor]):
+        B, L, D = x.shape
+        qkv = self.qkv(x).view(B, L, 3, self.n_heads, self.head_dim)
+        q, k, v = qkv.unbind(dim=2)                       # [B, L, H, dh]
+        q = q.permute(0, 2, 1, 3)                         # [B, H, L, dh]
+        k = k.permute(0, 2, 3, 1)                         # [B, H, dh, L]
+        v = v.permute(0, 2, 1, 3)
+        scores = torch.matmul(q, k)                       # [B, H, L, L]
+        p = ops.fused_softmax(scores, mask_bias, self.scale)
+        out = torch.matmul(p, v)                          # [B, H, L, dh]
+        out = out.permute(0, 2, 1, 3).reshape(B, L, D)
+        return self.proj(out)
+
+
+class FFN(nn.Module):
+    """d_model -> d_ff (bias+GeLU fused into the gfx950 kernel) -> d_model."""
+
+    def __init__(self, cfg: MLTCConfig):
+        super().__init__()
+        self.up = nn.Linear(cfg.d_model, cfg.d_ff, bias=False)
+        self.up_bias = nn.Parameter(torch.zeros(cfg.d_ff))
+        self.down = nn.Linear(cfg.d_ff, cfg.d_model, bias=True)
+
+    def forward(self, x):
+        h = self.up(x)
+        h = ops.fused_bias_gelu(h, self.up_bias)
+        return self.down(h)
+
+
+class EncoderBlock(nn.Module):
+    def __init__(self, cfg: MLTCConfig):
+        super().__init__()
+        self.ln1 = FusedLayerNorm(cfg.d_model, cfg.layer_norm_eps)
+        self.attn = Attention(cfg)
+        self.ln2 = FusedLayerNorm(cfg.d_model, cfg.layer_norm_eps)
+        self.ffn = FFN(cfg)
+        self.dropout = cfg.dropout
+
+    def forward(self, x, mask_bias):
+        h = self.attn(self.ln1(x), mask_bias)
+        if self.dropout:
+            h = F.dropout(h, self.dropout, self.training)
+        x = x + h
+        h = self.ffn(self.ln2(x))
+        if self.dropout:
+            h = F.dropout(h, self.dropout, self.training)
+        return x + h
+
+
+class MLTC(nn.Module):
+    def __init__(self, cfg: MLTCConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.d_model)
+        self.pos_emb = nn.Embedding(cfg.max_seq, cfg.d_model)
+        self.blocks = nn.ModuleList(EncoderBlock(cfg) for _ in range(cfg.n_layers))
+        self.ln_f = FusedLayerNorm(cfg.d_model, cfg.layer_norm_eps)
+        self.heads = nn.ModuleDict(
+            {name: nn.Linear(cfg.d_model, n) for name, n in cfg.heads.items()}
+        )
+        self.apply(self._init)
+
+    @staticmethod
+    def _init(m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=0.02)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
+
+    def forward(self, tokens: torch.Tensor,
+                attn_mask: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
+        """tokens: [B, L] int64; attn_mask: [B, L] bool (True = valid)."""
+        B, L = tokens.shape
+        pos = torch.arange(L, device=tokens.device)
+        x = self.tok_emb(tokens) + self.pos_emb(pos)[None, :, :]
+        mask_bias = None
+        if attn_mask is not None:
+            mask_bias = torch.where(
+                attn_mask, 0.0, -1e9
+            ).to(torch.float32).contiguous()
+        for blk in self.blocks:
+            x = blk(x, mask_bias)
+        x = self.ln_f(x)
+        # masked mean-pool over valid tokens
+        if attn_mask is not None:
+            w = attn_mask.to(x.dtype).unsqueeze(-1)
+            pooled = (x * w).sum(dim=1) / w.sum(dim=1).clamp(min=1)
+        else:
+            pooled = x.mean(dim=1)
+        return {name: head(pooled) for name, head in self.heads.items()}
+
+    def loss(self, logits: Dict[str, torch.Tensor],
+             labels: Dict[str, torch.Tensor]) -> torch.Tensor:
+        total = None
+        for name, lg in logits.items():
+            if name not in labels:
+                continue
+            lg = lg.float()
+            if name in MULTILABEL_HEADS:
+                li = F.binary_cross_entropy_with_logits(lg, labels[name].float())
+            else:
+                li = F.cross_entropy(lg, labels[name])
+            total = li if total is None else total + li
+        assert total is not None, "no labels matched any head"
+        return total
+
+
+def build_model(name_or_cfg, dtype=torch.bfloat16) -> MLTC:
+    cfg = CONFIGS[name_or_cfg] if isinstance(name_or_cfg, str) else name_or_cfg
+    model = MLTC(cfg)
+    return model.to(dtype)
