@@ -1,0 +1,122 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark: MLTC (mltc-base) classifier training step.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+For N > 1 the driver launches this under torch.distributed.run, one rank per
+GPU over RCCL.  W untimed warmup steps, then exactly K timed steps bracketed
+by barrier + torch.cuda.synchronize() on both sides; wall time is the MAX
+over ranks; rank 0 prints one JSON line.
+
+Metric: whole-job training throughput in tokens/s, bf16, synthetic data
+(random token ids + random taxonomy labels, random-init weights — there is no
+network for datasets).  The reference publishes no performance numbers
+(BASELINE.json: metric N/A, published {}), so vs_baseline is null.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=24)
+    ap.add_argument("--warmup", type=int, default=6)
+    ap.add_argument("--batch", type=int, default=64, help="per-GPU batch")
+    ap.add_argument("--seq", type=int, default=512)
+    ap.add_argument("--model", type=str, default="mltc-base")
+    ap.add_argument("--bucket-mb", type=int, default=64)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if world > 1:
+        import torch.distributed as dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        dist.init_process_group(
+            backend="nccl" if use_gpu else "gloo", rank=rank, world_size=world)
+
+    from tosem2021_amd.data.synthetic import synthetic_batch
+    from tosem2021_amd.models.classifier import CONFIGS
+    from tosem2021_amd.train import TrainConfig, Trainer
+
+    cfg = CONFIGS[args.model]
+    tcfg = TrainConfig(model=args.model, warmup_steps=0,
+                       total_steps=max(args.steps * 100, 1000),
+                       bucket_mb=args.bucket_mb)
+    trainer = Trainer(tcfg, device=device)
+
+    tokens, mask, labels = synthetic_batch(
+        cfg, args.batch, args.seq, device=device, seed=1234 + rank)
+
+    def barrier_sync():
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.step(tokens, mask, labels)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.step(tokens, mask, labels)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device
+                         if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if use_gpu else args.gpus
+    tokens_per_step = args.batch * args.seq * world
+    value = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_tokens_per_s",
+            "value": value,
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": args.seq,
+                "parallelism": f"dp{world}",
+            },
+        }))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

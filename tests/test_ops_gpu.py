@@ -1,0 +1,138 @@
+"""GPU numerics: each gfx950 HIP kernel vs the plain-PyTorch fp32 reference.
+
+Every test compares the hand-written kernel (bf16 storage, f32 math) against
+the same op computed in fp32 by reference.py; tolerances are bf16-rounding
+sized.  Marked gpu — the driver runs these on a real MI355X.
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from tosem2021_amd import ops
+from tosem2021_amd.ops import reference as ref
+
+
+def _bf16(x):
+    return x.to(torch.bfloat16).cuda().contiguous()
+
+
+@pytest.mark.parametrize("N,D", [(64, 1024), (257, 512), (32, 4096), (128, 136)])
+def test_layernorm_fwd(N, D):
+    torch.manual_seed(0)
+    x = torch.randn(N, D)
+    g = torch.randn(D)
+    b = torch.randn(D)
+    xg, gg, bg = _bf16(x), _bf16(g), _bf16(b)
+    y, mean, rstd = ops.hip_ops().layernorm_fwd(xg, gg, bg, 1e-5)
+    ye, me, re_ = ref.layernorm_fwd(xg.float().cpu(), gg.float().cpu(),
+                                    bg.float().cpu(), 1e-5)
+    assert torch.allclose(y.float().cpu(), ye, atol=3e-2, rtol=2e-2)
+    assert torch.allclose(mean.cpu(), me, atol=2e-3, rtol=1e-3)
+    assert torch.allclose(rstd.cpu(), re_, atol=2e-3, rtol=2e-3)
+
+
+@pytest.mark.parametrize("N,D", [(64, 1024), (257, 512), (32, 4096), (128, 136)])
+def test_layernorm_bwd(N, D):
+    torch.manual_seed(1)
+    xg = _bf16(torch.randn(N, D))
+    gg = _bf16(torch.randn(D))
+    bg = _bf16(torch.randn(D))
+    dyg = _bf16(torch.randn(N, D))
+    _, mean, rstd = ops.hip_ops().layernorm_fwd(xg, gg, bg, 1e-5)
+    dx, dgamma, dbeta = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd)
+    dxe, dge, dbe = ref.layernorm_bwd(dyg.float().cpu(), xg.float().cpu(),
+                                      gg.float().cpu(), mean.cpu(), rstd.cpu())
+    assert torch.allclose(dx.float().cpu(), dxe, atol=4e-2, rtol=3e-2)
+    tol = dict(atol=0.3 + 0.02 * math.sqrt(N), rtol=2e-2)
+    assert torch.allclose(dgamma.cpu(), dge, **tol)
+    assert torch.allclose(dbeta.cpu(), dbe, **tol)
+
+
+@pytest.mark.parametrize("N,D", [(128, 4096), (63, 256)])
+def test_bias_gelu(N, D):
+    torch.manual_seed(2)
+    xg = _bf16(torch.randn(N, D))
+    bg = _bf16(torch.randn(D))
+    y = ops.hip_ops().bias_gelu_fwd(xg, bg)
+    ye = ref.bias_gelu_fwd(xg.float().cpu(), bg.float().cpu())
+    assert torch.allclose(y.float().cpu(), ye, atol=3e-2, rtol=2e-2)
+    dyg = _bf16(torch.randn(N, D))
+    dx, dbias = ops.hip_ops().bias_gelu_bwd(dyg, xg, bg)
+    dxe, dbe = ref.bias_gelu_bwd(dyg.float().cpu(), xg.float().cpu(),
+                                 bg.float().cpu())
+    assert torch.allclose(dx.float().cpu(), dxe, atol=3e-2, rtol=2e-2)
+    assert torch.allclose(dbias.cpu(), dbe, atol=0.3 + 0.02 * math.sqrt(N),
+                          rtol=2e-2)
+
+
+@pytest.mark.parametrize("B,H,Lq,Lk", [(2, 4, 512, 512), (3, 2, 128, 128),
+                                       (2, 2, 64, 2056)])
+def test_softmax(B, H, Lq, Lk):
+    torch.manual_seed(3)
+    sg = _bf16(torch.randn(B, H, Lq, Lk) * 3)
+    mask = torch.zeros(B, Lk)
+    mask[:, Lk - Lk // 4:] = -1e9
+    mg = mask.cuda().contiguous()
+    scale = 1.0 / math.sqrt(64)
+    p = ops.hip_ops().softmax_fwd(sg, mg, scale)
+    pe = ref.softmax_fwd(sg.float().cpu(), mask, scale)
+    assert torch.allclose(p.float().cpu(), pe, atol=8e-3, rtol=2e-2)
+    assert float(p.float().cpu()[..., Lk - Lk // 4:].max()) < 1e-6
+    # row sums ~ 1
+    assert torch.allclose(p.float().sum(-1).cpu(), torch.ones(B, H, Lq),
+                          atol=3e-2)
+    dpg = _bf16(torch.randn(B, H, Lq, Lk))
+    ds = ops.hip_ops().softmax_bwd(dpg, p, scale)
+    dse = ref.softmax_bwd(dpg.float().cpu(), p.float().cpu(), scale)
+    assert torch.allclose(ds.float().cpu(), dse, atol=8e-3, rtol=2e-2)
+
+
+def test_softmax_no_mask():
+    torch.manual_seed(4)
+    sg = _bf16(torch.randn(2, 2, 64, 512))
+    p = ops.hip_ops().softmax_fwd(sg, None, 0.5)
+    pe = ref.softmax_fwd(sg.float().cpu(), None, 0.5)
+    assert torch.allclose(p.float().cpu(), pe, atol=8e-3, rtol=2e-2)
+
+
+def test_adamw_step():
+    torch.manual_seed(5)
+    n = 4096
+    master = torch.randn(n)
+    p = _bf16(master)
+    mst = master.cuda()
+    g = torch.randn(n)
+    gg = _bf16(g)
+    m = torch.zeros(n).cuda()
+    v = torch.zeros(n).cuda()
+    mr = master.clone()
+    pr = mr.to(torch.bfloat16)
+    mref = torch.zeros(n)
+    vref = torch.zeros(n)
+    for step in range(1, 5):
+        ops.hip_ops().adamw_step(p, gg, m, v, mst, 1e-2, 0.9, 0.999, 1e-8,
+                                 0.05, step, 1.0)
+        ref.adamw_step(pr, gg.float().cpu(), mref, vref, mr, lr=1e-2,
+                       beta1=0.9, beta2=0.999, eps=1e-8, wd=0.05, step=step)
+    assert torch.allclose(mst.cpu(), mr, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(p.float().cpu(), pr.float(), atol=1e-2)
+
+
+def test_grad_scale_matches_prescaled():
+    """grad_scale folding (the DDP average) == scaling the grad explicitly."""
+    torch.manual_seed(6)
+    n = 1024
+    master = torch.randn(n)
+    g = torch.randn(n)
+    pa = _bf16(master); ma = master.cuda().clone()
+    sa = torch.zeros(n).cuda(); va = torch.zeros(n).cuda()
+    pb = _bf16(master); mb = master.cuda().clone()
+    sb = torch.zeros(n).cuda(); vb = torch.zeros(n).cuda()
+    ops.hip_ops().adamw_step(pa, _bf16(g), sa, va, ma, 1e-2, 0.9, 0.999,
+                             1e-8, 0.0, 1, 0.25)
+    ops.hip_ops().adamw_step(pb, _bf16(g * 0.25), sb, vb, mb, 1e-2, 0.9,
+                             0.999, 1e-8, 0.0, 1, 1.0)
+    assert torch.allclose(ma, mb, atol=2e-3, rtol=1e-3)

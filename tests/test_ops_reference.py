@@ -1,0 +1,86 @@
+"""CPU tests: the fp32 reference ops agree with plain PyTorch."""
+import math
+
+import pytest
+import torch
+
+from tosem2021_amd.ops import reference as ref
+
+
+def test_layernorm_fwd_matches_torch():
+    torch.manual_seed(0)
+    x = torch.randn(8, 64)
+    g = torch.randn(64)
+    b = torch.randn(64)
+    y, mean, rstd = ref.layernorm_fwd(x, g, b, 1e-5)
+    expect = torch.nn.functional.layer_norm(x, (64,), g, b, 1e-5)
+    assert torch.allclose(y, expect, atol=1e-5)
+    assert torch.allclose(mean, x.mean(-1).reshape(-1), atol=1e-6)
+
+
+def test_layernorm_bwd_matches_autograd():
+    torch.manual_seed(1)
+    x = torch.randn(6, 32, requires_grad=True)
+    g = torch.randn(32, requires_grad=True)
+    b = torch.randn(32, requires_grad=True)
+    y = torch.nn.functional.layer_norm(x, (32,), g, b, 1e-5)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    _, mean, rstd = ref.layernorm_fwd(x.detach(), g.detach(), b.detach(), 1e-5)
+    dx, dg, db = ref.layernorm_bwd(dy, x.detach(), g.detach(), mean, rstd)
+    assert torch.allclose(dx, x.grad, atol=1e-4)
+    assert torch.allclose(dg, g.grad, atol=1e-4)
+    assert torch.allclose(db, b.grad, atol=1e-4)
+
+
+def test_bias_gelu_matches_torch():
+    torch.manual_seed(2)
+    x = torch.randn(4, 16)
+    b = torch.randn(16)
+    y = ref.bias_gelu_fwd(x, b)
+    expect = torch.nn.functional.gelu(x + b, approximate="tanh")
+    assert torch.allclose(y, expect, atol=1e-5)
+    xx = (x + b).detach().requires_grad_(True)
+    torch.nn.functional.gelu(xx, approximate="tanh").backward(torch.ones_like(x))
+    dx, dbias = ref.bias_gelu_bwd(torch.ones_like(x), x, b)
+    assert torch.allclose(dx, xx.grad, atol=1e-5)
+    assert torch.allclose(dbias, xx.grad.sum(0), atol=1e-5)
+
+
+def test_softmax_matches_torch():
+    torch.manual_seed(3)
+    s = torch.randn(2, 3, 4, 8)
+    mask = torch.zeros(2, 8)
+    mask[0, 5:] = -1e9
+    scale = 1.0 / math.sqrt(16)
+    p = ref.softmax_fwd(s, mask, scale)
+    expect = torch.softmax(s * scale + mask.view(2, 1, 1, 8), dim=-1)
+    assert torch.allclose(p, expect, atol=1e-5)
+    assert torch.all(p[0, :, :, 5:] < 1e-6)
+    # bwd against autograd
+    s2 = s.clone().requires_grad_(True)
+    p2 = torch.softmax(s2 * scale + mask.view(2, 1, 1, 8), dim=-1)
+    dp = torch.randn_like(p2)
+    p2.backward(dp)
+    ds = ref.softmax_bwd(dp, p.detach(), scale)
+    assert torch.allclose(ds, s2.grad, atol=1e-5)
+
+
+def test_adamw_matches_torch_optimizer():
+    torch.manual_seed(4)
+    n = 64
+    master = torch.randn(n)
+    p = master.clone()
+    g = torch.randn(n)
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    # torch reference
+    tp = master.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([tp], lr=1e-2, betas=(0.9, 0.999), eps=1e-8,
+                            weight_decay=0.05)
+    for step in range(1, 4):
+        tp.grad = g.clone()
+        opt.step()
+        ref.adamw_step(p, g, m, v, master, lr=1e-2, beta1=0.9, beta2=0.999,
+                       eps=1e-8, wd=0.05, step=step)
+    assert torch.allclose(master, tp.detach(), atol=1e-5)
