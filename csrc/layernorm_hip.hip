@@ -192,7 +192,7 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
         for (int j = 0; j < 8; ++j) {
           float d = bf16_to_f32(vd[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
-          float g = d * gamma[i + j];
+          float g = d * bf16_to_f32(gamma[i + j]);
           o[j] = f32_to_bf16(rstd * (g - s1 - h * s2));
           atomicAdd(&sg[i + j], d * h);
           atomicAdd(&sb[i + j], d);
