@@ -1,0 +1,135 @@
+"""Unit tests: extractors on synthetic sources (SURVEY.md §4 test model —
+unit layer: parsers/classifiers on synthetic test files)."""
+import os
+import textwrap
+
+from tosem2021_amd.corpus import walker
+from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+from tosem2021_amd.extract.python_extractor import extract_file
+from tosem2021_amd.extract.ts_extractor import extract_ts_file
+
+PY_SRC = textwrap.dedent('''
+    import pytest
+    from unittest import mock
+
+    class TestThing:
+        def test_accuracy(self):
+            """checks model accuracy"""
+            score = 0.97
+            self.assertAlmostEqual(0.96, score, places=2)
+            assert score > 0.8
+
+        def test_raises(self):
+            with pytest.raises(ValueError):
+                parse("bad")
+
+    @pytest.mark.parametrize("x", [1, 2])
+    def test_mocked(x, monkeypatch):
+        m = mock.MagicMock()
+        assert isinstance(m, object)
+
+    def helper():
+        assert False  # not a test
+''')
+
+CC_SRC = textwrap.dedent('''
+    #include "gtest/gtest.h"
+    TEST(MathTest, HandlesZero) {
+      EXPECT_EQ(0, Add(0, 0));
+      EXPECT_NEAR(1.0, Compute(), 1e-6);
+    }
+    TEST_F(PipelineFixture, Throws) {
+      EXPECT_THROW(Run("bad"), std::invalid_argument);
+      if (x) { EXPECT_TRUE(ok()); }
+    }
+''')
+
+TS_SRC = textwrap.dedent('''
+    describe('manager', () => {
+        it('starts an experiment', async () => {
+            const r = await mgr.start();
+            expect(r.status).to.equal('RUNNING');
+            assert.strictEqual(r.id, 1);
+        });
+        it('rejects bad config', () => {
+            expect(() => mgr.load('x')).to.throw();
+        });
+    });
+''')
+
+
+def _write(tmp_path, name, content):
+    p = tmp_path / name
+    p.write_text(content)
+    return str(p)
+
+
+def test_python_extractor(tmp_path):
+    cases = extract_file(_write(tmp_path, "test_thing.py", PY_SRC))
+    names = {c.qualname for c in cases}
+    assert names == {"TestThing.test_accuracy", "TestThing.test_raises",
+                     "test_mocked"}
+    acc = next(c for c in cases if c.name == "test_accuracy")
+    assert acc.docstring == "checks model accuracy"
+    kinds = {a.kind for a in acc.assertions}
+    assert "unittest" in kinds and "assert" in kinds
+    raises = next(c for c in cases if c.name == "test_raises")
+    assert any(a.kind == "raises" and a.exception == "ValueError"
+               for a in raises.assertions)
+    mocked = next(c for c in cases if c.name == "test_mocked")
+    assert mocked.uses_mock and mocked.is_parametrized
+
+
+def test_python_extractor_handles_syntax_error(tmp_path):
+    assert extract_file(_write(tmp_path, "test_py2.py",
+                               "print 'hello'\ndef test_x(): pass")) == []
+
+
+def test_gtest_extractor(tmp_path):
+    cases = extract_gtest_file(_write(tmp_path, "math_test.cc", CC_SRC))
+    assert {c.qualname for c in cases} == {"MathTest.HandlesZero",
+                                           "PipelineFixture.Throws"}
+    zero = next(c for c in cases if c.name == "HandlesZero")
+    assert len(zero.assertions) == 2
+    assert any(a.kind == "approx" for a in zero.assertions)
+    throws = next(c for c in cases if c.name == "Throws")
+    kinds = [a.kind for a in throws.assertions]
+    assert "raises" in kinds and "unittest" in kinds
+    exc = next(a for a in throws.assertions if a.kind == "raises")
+    assert "invalid_argument" in exc.exception
+
+
+def test_ts_extractor(tmp_path):
+    cases = extract_ts_file(_write(tmp_path, "manager.test.ts", TS_SRC))
+    assert len(cases) == 2
+    first = cases[0]
+    assert first.qualname == "manager.starts an experiment"
+    assert len(first.assertions) == 2
+    assert any(a.kind == "raises" for a in cases[1].assertions)
+
+
+def test_walker_detection(tmp_path):
+    (tmp_path / "pkg").mkdir()
+    (tmp_path / "tests").mkdir()
+    files = {
+        "pkg/module.py": "x = 1",
+        "pkg/module_test.py": "def test_a(): pass",
+        "tests/test_b.py": "def test_b(): pass",
+        "tests/helper.py": "pass",
+        "pkg/engine_test.cc": "TEST(A,B){}",
+        "pkg/engine.cc": "",
+        "pkg/ui.test.ts": "it('x',()=>{});",
+        "pkg/ui.ts": "",
+    }
+    for rel, content in files.items():
+        p = tmp_path / rel
+        p.write_text(content)
+    found = {f.rel for f in walker.test_files(str(tmp_path))}
+    assert "pkg/module_test.py" in found
+    assert "tests/test_b.py" in found
+    assert "tests/helper.py" in found          # in a tests/ dir
+    assert "pkg/engine_test.cc" in found
+    assert "pkg/ui.test.ts" in found
+    assert "pkg/module.py" not in found
+    assert "pkg/engine.cc" not in found
+    assert "pkg/ui.ts" not in found

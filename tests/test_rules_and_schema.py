@@ -1,0 +1,97 @@
+"""Unit tests: rule classifier + taxonomy schema."""
+from tosem2021_amd.classify.rules import classify_stage, classify_text
+from tosem2021_amd.extract.schema import (
+    PROPERTIES, STAGES, STRATEGIES, TAXONOMY_COLUMNS, TestCaseRow,
+    canonical_property)
+
+
+def test_taxonomy_has_41_columns():
+    assert len(TAXONOMY_COLUMNS) == 41
+    assert TAXONOMY_COLUMNS[0] == "Index"
+    assert TAXONOMY_COLUMNS[-1] == "None_above"
+
+
+def test_label_space_sizes():
+    assert len(STRATEGIES) == 19
+    assert len(PROPERTIES) == 21
+    assert len(STAGES) == 9
+
+
+def test_csv_row_roundtrip_width():
+    row = classify_text("assertAlmostEqual(0.96, accuracy_score(p, y))")
+    assert len(row.to_csv_row()) == len(TAXONOMY_COLUMNS)
+
+
+def test_rounding_tolerance_rule():
+    row = classify_text("assertAlmostEqual(0.96, accuracy_score(pred, y))")
+    assert row.flags.get("Approximation") == 1
+    assert row.approximation_type == "rounding_tolence"
+    assert "rounding_tolence" in row.strategies()
+
+
+def test_allclose_rule():
+    row = classify_text("np.testing.assert_allclose(a, b, rtol=1e-5)")
+    assert "absolute_relative_tolerence" in row.strategies()
+
+
+def test_raises_value_error():
+    row = classify_text("with pytest.raises(ValueError): f(-1)")
+    assert row.flags.get("error_handling") == 1
+    assert row.error_type == "ValueError"
+    assert "value_error" in row.strategies()
+    assert "negative_test" in row.strategies()
+
+
+def test_instance_and_subset_checks():
+    assert "instance_check" in classify_text(
+        "assert isinstance(clf, BaseEstimator)").strategies()
+    assert "sub_set_checks" in classify_text(
+        "assertIn(k, results.keys())").strategies()
+
+
+def test_null_and_status():
+    assert "Null_pointer" in classify_text("assert x is not None").strategies()
+    assert "status_analysis" in classify_text(
+        "assertTrue(job.is_running())").strategies()
+
+
+def test_gtest_assertions_classify():
+    row = classify_text("EXPECT_NEAR(1.0, Compute(), 1e-6)")
+    assert row.flags.get("Approximation") == 1
+    row = classify_text("EXPECT_EQ(nullptr, head)")
+    assert "Null_pointer" in row.strategies()
+
+
+def test_method_from_path():
+    row = classify_text("assert x == 1", path="tests/integration_tests/test_a.py")
+    assert row.method == "integration"
+    row = classify_text("assert x == 1", path="tests/regression/test_b.py")
+    assert row.method == "regression"
+    row = classify_text("assert ok", name="test_end_to_end_flow")
+    assert row.method == "end_to_end"
+    row = classify_text("assert ok", path="tests/unit/test_c.py")
+    assert row.method == "unit_test"
+
+
+def test_stage_classification():
+    assert classify_stage("download dataset loader") == "data_collection"
+    assert classify_stage("train the model optimizer") == "model_training"
+    assert classify_stage("parse config flags") == "config_utility"
+    assert classify_stage("no cues at all zzz") == "config_utility"
+
+
+def test_canonical_property():
+    assert canonical_property("Roboustness") == "Robustness"
+    assert canonical_property("Fault Tolerance") == "Robustness"
+    assert canonical_property("Computing Efficiency") == "Efficiency"
+    assert canonical_property("Validity") == "Data Validity"
+    assert canonical_property(None) is None
+    assert canonical_property(float("nan")) is None
+
+
+def test_testcaserow_method_precedence():
+    r = TestCaseRow(flags={"Integration": 1, "regression": 1})
+    assert r.method == "integration"
+    r = TestCaseRow(flags={"end_to_end": 1, "Integration": 1})
+    assert r.method == "end_to_end"
+    assert TestCaseRow(flags={}).method == "unit_test"

@@ -1,0 +1,70 @@
+"""Tokenizer determinism + neural classifier training entry (CPU, tiny)."""
+import csv
+import os
+
+import torch
+
+from tosem2021_amd.extract.schema import TAXONOMY_COLUMNS
+from tosem2021_amd.models.tokenizer import CLS, PAD, CodeTokenizer
+
+
+def test_tokenizer_deterministic_and_bounded():
+    tok = CodeTokenizer(512)
+    ids1 = tok.encode("assertAlmostEqual(0.96, accuracy_score(p, y))", 64)
+    ids2 = tok.encode("assertAlmostEqual(0.96, accuracy_score(p, y))", 64)
+    assert ids1 == ids2
+    assert ids1[0] == CLS
+    assert all(0 <= i < 512 for i in ids1)
+    # camelCase / snake_case split to shared subwords
+    t1 = tok.tokens("assertAlmostEqual")
+    t2 = tok.tokens("assert_almost_equal")
+    assert t1 == t2 == ["assert", "almost", "equal"]
+
+
+def test_encode_batch_mask():
+    tok = CodeTokenizer(512)
+    toks, mask = tok.encode_batch(["a b c", "x"], 16)
+    assert toks.shape == mask.shape
+    assert bool(mask[0].sum() > mask[1].sum())
+    assert toks[1, mask[1].sum():].eq(PAD).all()
+
+
+def _write_tiny_taxonomy(path, n=48):
+    rows = []
+    for i in range(n):
+        base = {c: "" for c in TAXONOMY_COLUMNS}
+        for c in ("regression", "Integration", "end_to_end", "status_test",
+                  "negative_test", "value_range", "null_pointer",
+                  "logical_statement", "logical_expression", "error_handling",
+                  "Approximation", "basic_comparizon"):
+            base[c] = 0
+        if i % 2:
+            base.update(Labels=f"assertTrue(ok_{i}())", status_test=1,
+                        Category="Model", Repo="Ray", Model="Correctness")
+        else:
+            base.update(Labels=f"assertAlmostEqual(a_{i}, b)",
+                        Approximation=1,
+                        Approximation_Type="rounding_tolence",
+                        Category="Data Preprocessing", Repo="tpot",
+                        Data="Validity")
+        base["Index"] = i
+        rows.append(base)
+    with open(path, "w", newline="") as f:
+        w = csv.DictWriter(f, fieldnames=TAXONOMY_COLUMNS)
+        w.writeheader()
+        w.writerows(rows)
+
+
+def test_train_classifier_tiny_cpu(tmp_path):
+    from tosem2021_amd.classify.neural import train_classifier
+    tax = str(tmp_path / "tiny_tax.csv")
+    _write_tiny_taxonomy(tax)
+    res = train_classifier(tax, model="mltc-tiny", steps=6, batch=8, seq=32,
+                           lr=1e-3, device="cpu",
+                           ckpt_dir=str(tmp_path / "ck"))
+    assert res["steps"] == 6
+    assert res["final_loss"] == res["final_loss"]  # not NaN
+    for k in ("strategy_micro_f1", "property_micro_f1", "method_accuracy"):
+        assert k in res
+    # checkpoint written
+    assert any(f.startswith("ckpt_") for f in os.listdir(tmp_path / "ck"))

@@ -1,0 +1,118 @@
+"""Learned taxonomy classifier: train/evaluate MLTC on labeled taxonomy rows.
+
+The one place the reference's capability legitimately meets the GPU
+(SURVEY.md §7 layer 3): fine-tune the transformer classifier on the study's
+9,685 human-labeled rows and beat the rule engine's agreement scores.
+"""
+from __future__ import annotations
+
+import math
+import os
+import time
+from typing import Dict, Optional
+
+import torch
+
+from tosem2021_amd.analyze.taxonomy import load_taxonomy
+from tosem2021_amd.classify.agreement import LabelScore, micro_f1
+from tosem2021_amd.data.dataset import TaxonomyDataset
+from tosem2021_amd.extract.schema import METHODS, PROPERTIES, STRATEGIES
+from tosem2021_amd.models.classifier import CONFIGS, MLTCConfig
+from tosem2021_amd.models.tokenizer import CodeTokenizer
+from tosem2021_amd.train import TrainConfig, Trainer
+
+
+@torch.no_grad()
+def evaluate(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
+             seq: int, batch: int = 64, threshold: float = 0.5) -> Dict[str, float]:
+    model = trainer.model
+    model.eval()
+    device = trainer.device
+    strat_scores = {l: LabelScore(l) for l in STRATEGIES}
+    prop_scores = {l: LabelScore(l) for l in PROPERTIES}
+    method_hits = 0
+    n = 0
+    for toks, mask, labels in ds.batches(tok, batch, seq, device=device,
+                                         shuffle=False):
+        logits = model(toks, mask)
+        sp = (torch.sigmoid(logits["strategy"].float()) > threshold)
+        pp = (torch.sigmoid(logits["property"].float()) > threshold)
+        mp = logits["method"].float().argmax(-1)
+        gs = labels["strategy"] > 0.5
+        gp = labels["property"] > 0.5
+        for i, l in enumerate(STRATEGIES):
+            s = strat_scores[l]
+            s.tp += int((sp[:, i] & gs[:, i]).sum())
+            s.fp += int((sp[:, i] & ~gs[:, i]).sum())
+            s.fn += int((~sp[:, i] & gs[:, i]).sum())
+        for i, l in enumerate(PROPERTIES):
+            s = prop_scores[l]
+            s.tp += int((pp[:, i] & gp[:, i]).sum())
+            s.fp += int((pp[:, i] & ~gp[:, i]).sum())
+            s.fn += int((~pp[:, i] & gp[:, i]).sum())
+        method_hits += int((mp == labels["method"]).sum())
+        n += toks.shape[0]
+    model.train()
+    return {
+        "strategy_micro_f1": micro_f1(strat_scores),
+        "property_micro_f1": micro_f1(prop_scores),
+        "method_accuracy": method_hits / max(n, 1),
+        "n_eval": n,
+    }
+
+
+def train_classifier(taxonomy_path: str, model: str = "mltc-base",
+                     steps: int = 500, batch: int = 32, seq: int = 256,
+                     lr: float = 3e-4, ckpt_dir: Optional[str] = None,
+                     resume: bool = False, device: Optional[str] = None,
+                     eval_every: int = 0, seed: int = 0) -> dict:
+    dev = torch.device(device) if device else (
+        torch.device("cuda") if torch.cuda.is_available() else
+        torch.device("cpu"))
+    df = load_taxonomy(taxonomy_path)
+    full = TaxonomyDataset.from_taxonomy(df)
+    train_ds, val_ds = full.split(val_frac=0.1, seed=seed)
+
+    base = CONFIGS[model]
+    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
+    tok = CodeTokenizer(cfg.vocab_size)
+    tcfg = TrainConfig(model=model, lr=lr, warmup_steps=min(50, steps // 10),
+                       total_steps=steps, ckpt_dir=ckpt_dir,
+                       dtype="bf16" if dev.type == "cuda" else "f32")
+    trainer = Trainer(tcfg, device=dev, model_cfg=cfg)
+    if resume and ckpt_dir:
+        trainer.load_or_init()
+
+    t0 = time.time()
+    losses = []
+    epoch = 0
+    it = iter(())
+    while trainer.step_num < steps:
+        try:
+            toks, mask, labels = next(it)
+        except StopIteration:
+            it = train_ds.batches(tok, batch, seq, device=dev, shuffle=True,
+                                  seed=seed + epoch, drop_last=True)
+            epoch += 1
+            continue
+        losses.append(trainer.step(toks, mask, labels))
+        if eval_every and trainer.step_num % eval_every == 0:
+            ev = evaluate(trainer, val_ds, tok, seq)
+            print(f"step {trainer.step_num} loss {losses[-1]:.4f} "
+                  f"strategyF1 {ev['strategy_micro_f1']:.3f} "
+                  f"propF1 {ev['property_micro_f1']:.3f} "
+                  f"methodAcc {ev['method_accuracy']:.3f}")
+    train_time = time.time() - t0
+    ev = evaluate(trainer, val_ds, tok, seq)
+    if ckpt_dir:
+        trainer.save()
+    return {
+        "steps": trainer.step_num,
+        "epochs": epoch,
+        "train_time_s": round(train_time, 2),
+        "final_loss": losses[-1] if losses else None,
+        "loss_first10_mean": (sum(losses[:10]) / min(len(losses), 10)
+                              if losses else None),
+        **{k: (round(v, 4) if isinstance(v, float) else v)
+           for k, v in ev.items()},
+    }

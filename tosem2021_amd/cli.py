@@ -1,0 +1,127 @@
+"""Command-line interface: survey-style pipeline commands.
+
+    python -m tosem2021_amd.cli mine --projects auto-sklearn --out tax.csv
+    python -m tosem2021_amd.cli analyze --taxonomy tax.csv --out out/RQs
+    python -m tosem2021_amd.cli agreement --taxonomy /root/reference/RQs/taxonomy_test2.csv
+    python -m tosem2021_amd.cli golden --ours out/RQs --reference /root/reference/RQs
+    python -m tosem2021_amd.cli train --taxonomy ... --steps 200
+    python -m tosem2021_amd.cli report --taxonomy tax.csv
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+
+def cmd_mine(args):
+    from tosem2021_amd.corpus.registry import PROJECTS, available_projects
+    from tosem2021_amd.pipeline import mine
+    projects = args.projects or available_projects(args.corpus_root)
+    bad = [p for p in projects if p not in PROJECTS]
+    if bad:
+        sys.exit(f"unknown projects: {bad}; known: {list(PROJECTS)}")
+    path = mine(projects, args.out, corpus_root=args.corpus_root,
+                languages=tuple(args.languages.split(",")),
+                workers=args.workers)
+    print(f"wrote {path}")
+
+
+def cmd_analyze(args):
+    from tosem2021_amd.analyze.tables import write_all
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    df = load_taxonomy(args.taxonomy)
+    paths = write_all(df, args.out)
+    if not args.no_figures:
+        from tosem2021_amd.analyze.figures import write_figures
+        paths.update(write_figures(df, args.out))
+    for k, p in paths.items():
+        print(f"{k}: {p}")
+
+
+def cmd_agreement(args):
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    from tosem2021_amd.classify.agreement import (
+        evaluate_rules_on_taxonomy, report)
+    df = load_taxonomy(args.taxonomy)
+    res = evaluate_rules_on_taxonomy(df, limit=args.limit)
+    print(report(res))
+    if args.json:
+        with open(args.json, "w") as f:
+            json.dump(res, f, indent=2)
+
+
+def cmd_golden(args):
+    from tosem2021_amd.analyze.golden import golden_diff
+    res = golden_diff(args.ours, args.reference)
+    print(json.dumps(res, indent=2))
+    sys.exit(0 if res["ok"] else 1)
+
+
+def cmd_report(args):
+    from tosem2021_amd.analyze.report import summary_report
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    df = load_taxonomy(args.taxonomy)
+    print(summary_report(df))
+
+
+def cmd_train(args):
+    from tosem2021_amd.classify.neural import train_classifier
+    res = train_classifier(
+        taxonomy_path=args.taxonomy, model=args.model, steps=args.steps,
+        batch=args.batch, seq=args.seq, lr=args.lr, ckpt_dir=args.ckpt_dir,
+        resume=args.resume)
+    print(json.dumps(res, indent=2))
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="tosem2021_amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    p = sub.add_parser("mine", help="mine a corpus into a taxonomy CSV")
+    p.add_argument("--projects", nargs="*", default=None)
+    p.add_argument("--corpus-root", default=None)
+    p.add_argument("--languages", default="python,cpp,ts")
+    p.add_argument("--workers", type=int, default=0)
+    p.add_argument("--out", required=True)
+    p.set_defaults(fn=cmd_mine)
+
+    p = sub.add_parser("analyze", help="regenerate RQ1/RQ3/RQ4 tables+figures")
+    p.add_argument("--taxonomy", required=True)
+    p.add_argument("--out", required=True)
+    p.add_argument("--no-figures", action="store_true")
+    p.set_defaults(fn=cmd_analyze)
+
+    p = sub.add_parser("agreement", help="score rules vs the reference labels")
+    p.add_argument("--taxonomy", required=True)
+    p.add_argument("--limit", type=int, default=0)
+    p.add_argument("--json", default=None)
+    p.set_defaults(fn=cmd_agreement)
+
+    p = sub.add_parser("golden", help="diff regenerated tables vs reference")
+    p.add_argument("--ours", required=True)
+    p.add_argument("--reference", required=True)
+    p.set_defaults(fn=cmd_golden)
+
+    p = sub.add_parser("report", help="print a taxonomy summary report")
+    p.add_argument("--taxonomy", required=True)
+    p.set_defaults(fn=cmd_report)
+
+    p = sub.add_parser("train", help="train the MLTC classifier on a taxonomy")
+    p.add_argument("--taxonomy", required=True)
+    p.add_argument("--model", default="mltc-base")
+    p.add_argument("--steps", type=int, default=500)
+    p.add_argument("--batch", type=int, default=32)
+    p.add_argument("--seq", type=int, default=256)
+    p.add_argument("--lr", type=float, default=3e-4)
+    p.add_argument("--ckpt-dir", default=None)
+    p.add_argument("--resume", action="store_true")
+    p.set_defaults(fn=cmd_train)
+
+    args = ap.parse_args(argv)
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()

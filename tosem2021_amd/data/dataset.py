@@ -1,0 +1,87 @@
+"""Taxonomy dataset: the study's 9,685 labeled rows as classifier training
+data (tokens + multi-hot strategy/property labels + stage/method classes)."""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, Iterator, List, Optional, Tuple
+
+import pandas as pd
+import torch
+
+from tosem2021_amd.analyze.taxonomy import (
+    row_method, row_properties, row_stage, row_strategies)
+from tosem2021_amd.extract.schema import (
+    METHODS, PROPERTIES, STAGES, STRATEGIES)
+from tosem2021_amd.models.tokenizer import CodeTokenizer
+
+
+@dataclass
+class TaxonomyDataset:
+    texts: List[str]
+    strategy: torch.Tensor   # [N, 19] multi-hot
+    property_: torch.Tensor  # [N, 21] multi-hot
+    stage: torch.Tensor      # [N] class id
+    method: torch.Tensor     # [N] class id
+
+    def __len__(self) -> int:
+        return len(self.texts)
+
+    @classmethod
+    def from_taxonomy(cls, df: pd.DataFrame) -> "TaxonomyDataset":
+        strat_sets = row_strategies(df)
+        prop_sets = row_properties(df)
+        stages = row_stage(df).tolist()
+        methods = row_method(df).tolist()
+        n = len(df)
+        strategy = torch.zeros(n, len(STRATEGIES))
+        property_ = torch.zeros(n, len(PROPERTIES))
+        stage = torch.zeros(n, dtype=torch.long)
+        method = torch.zeros(n, dtype=torch.long)
+        for i in range(n):
+            for s in strat_sets[i]:
+                strategy[i, STRATEGIES.index(s)] = 1
+            for p in prop_sets[i]:
+                property_[i, PROPERTIES.index(p)] = 1
+            stage[i] = STAGES.index(stages[i]) if stages[i] in STAGES else \
+                STAGES.index("config_utility")
+            method[i] = METHODS.index(methods[i])
+        texts = (df["Labels"].astype(str) + " | " +
+                 df["Component"].astype(str)).tolist()
+        return cls(texts, strategy, property_, stage, method)
+
+    def split(self, val_frac: float = 0.1, seed: int = 0
+              ) -> Tuple["TaxonomyDataset", "TaxonomyDataset"]:
+        g = torch.Generator().manual_seed(seed)
+        perm = torch.randperm(len(self), generator=g)
+        n_val = int(len(self) * val_frac)
+        va, tr = perm[:n_val], perm[n_val:]
+
+        def take(ix):
+            return TaxonomyDataset(
+                [self.texts[i] for i in ix.tolist()],
+                self.strategy[ix], self.property_[ix],
+                self.stage[ix], self.method[ix])
+        return take(tr), take(va)
+
+    def batches(self, tokenizer: CodeTokenizer, batch_size: int, max_len: int,
+                device="cpu", shuffle: bool = True, seed: int = 0,
+                drop_last: bool = False
+                ) -> Iterator[Tuple[torch.Tensor, torch.Tensor,
+                                    Dict[str, torch.Tensor]]]:
+        n = len(self)
+        g = torch.Generator().manual_seed(seed)
+        order = torch.randperm(n, generator=g) if shuffle else torch.arange(n)
+        for lo in range(0, n, batch_size):
+            ix = order[lo:lo + batch_size]
+            if drop_last and len(ix) < batch_size:
+                break
+            toks, mask = tokenizer.encode_batch(
+                [self.texts[i] for i in ix.tolist()], max_len, device=device)
+            labels = {
+                "strategy": self.strategy[ix].to(device),
+                "property": self.property_[ix].to(device),
+                "stage": self.stage[ix].to(device),
+                "method": self.method[ix].to(device),
+            }
+            yield toks, mask, labels

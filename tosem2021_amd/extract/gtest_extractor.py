@@ -1,0 +1,125 @@
+"""C++ gtest extractor (regex + brace matching; no libclang in this image).
+
+Covers the corpus' gtest conventions at Apollo/Ray/DeepSpeech scale
+(665 apollo *_test.cc, 56 ray, 28 DeepSpeech):
+TEST / TEST_F / TEST_P / TYPED_TEST / TYPED_TEST_P macros and the
+EXPECT_* / ASSERT_* assertion families.
+"""
+from __future__ import annotations
+
+import re
+from typing import List, Optional
+
+from tosem2021_amd.extract.python_extractor import Assertion, TestCase
+
+RE_TEST_MACRO = re.compile(
+    r"\b(TEST|TEST_F|TEST_P|TYPED_TEST|TYPED_TEST_P|INSTANTIATE_TEST_SUITE_P)"
+    r"\s*\(\s*([A-Za-z_]\w*)\s*,\s*([A-Za-z_]\w*)\s*\)")
+RE_ASSERT = re.compile(
+    r"\b((?:EXPECT|ASSERT)_[A-Z_0-9]+)\s*\(")
+RE_DEATH = re.compile(r"DEATH|THROW", re.I)
+
+
+def _match_brace_block(text: str, open_idx: int) -> int:
+    """Index just past the matching '}' for the '{' at open_idx (best-effort:
+    brace counting skipping string/char literals and comments)."""
+    depth = 0
+    i = open_idx
+    n = len(text)
+    while i < n:
+        c = text[i]
+        if c == '"' or c == "'":
+            q = c
+            i += 1
+            while i < n and text[i] != q:
+                if text[i] == "\\":
+                    i += 1
+                i += 1
+        elif c == "/" and i + 1 < n and text[i + 1] == "/":
+            i = text.find("\n", i)
+            if i < 0:
+                return n
+        elif c == "/" and i + 1 < n and text[i + 1] == "*":
+            i = text.find("*/", i)
+            if i < 0:
+                return n
+            i += 1
+        elif c == "{":
+            depth += 1
+        elif c == "}":
+            depth -= 1
+            if depth == 0:
+                return i + 1
+        i += 1
+    return n
+
+
+def _extract_call(text: str, start: int) -> str:
+    """Source of a call starting at `start` through its closing paren."""
+    depth = 0
+    i = start
+    n = len(text)
+    while i < n:
+        c = text[i]
+        if c == '"':
+            i += 1
+            while i < n and text[i] != '"':
+                if text[i] == "\\":
+                    i += 1
+                i += 1
+        elif c == "(":
+            depth += 1
+        elif c == ")":
+            depth -= 1
+            if depth == 0:
+                return text[start:i + 1]
+        i += 1
+    return text[start:min(start + 200, n)]
+
+
+def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
+    rel = rel or path
+    try:
+        with open(path, "rb") as f:
+            text = f.read().decode("utf-8", errors="replace")
+    except OSError:
+        return []
+    cases: List[TestCase] = []
+    for m in RE_TEST_MACRO.finditer(text):
+        macro, suite, name = m.group(1), m.group(2), m.group(3)
+        if macro == "INSTANTIATE_TEST_SUITE_P":
+            continue
+        brace = text.find("{", m.end())
+        if brace < 0:
+            continue
+        end = _match_brace_block(text, brace)
+        body = text[m.start():end]
+        lineno = text.count("\n", 0, m.start()) + 1
+        assertions: List[Assertion] = []
+        for am in RE_ASSERT.finditer(body):
+            call = am.group(1)
+            src = call + _extract_call(body, am.end() - 1)
+            a_line = lineno + body.count("\n", 0, am.start())
+            kind = "raises" if RE_DEATH.search(call) else "unittest"
+            exc = ""
+            if kind == "raises":
+                inner = src[src.find("(") + 1:]
+                parts = inner.rsplit(",", 1)
+                exc = parts[-1].strip(" );") if len(parts) > 1 else ""
+            if "NEAR" in call or "FLOAT_EQ" in call or "DOUBLE_EQ" in call:
+                kind = "approx"
+            assertions.append(Assertion(
+                kind=kind, call_name=call, source=src[:500], lineno=a_line,
+                exception=exc))
+        cases.append(TestCase(
+            name=name,
+            qualname=f"{suite}.{name}",
+            file_rel=rel,
+            lineno=lineno,
+            end_lineno=lineno + body.count("\n"),
+            source=body[:4000],
+            assertions=assertions,
+            uses_mock=bool(re.search(r"\bMOCK_METHOD|NiceMock|StrictMock|gmock",
+                                     body)),
+        ))
+    return cases
