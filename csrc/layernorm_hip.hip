@@ -126,14 +126,21 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   __syncthreads();
 
   const int pkts = D / (WAVE * 8);
+  // bwd caches 3 f32 arrays per lane -> cap at 4 packets (D <= 2048).
+  // Each (wave,lane,j) owns a FIXED column set across its whole row loop, so
+  // dgamma/dbeta partials accumulate in registers and hit the LDS once per
+  // wave at the end (the per-element LDS atomics were 8.5 ms/step in the
+  // baseline profile — profiles/r01_kernel_stats_baseline.md).
+#define MAX_PKT_BWD 4
+  float dg_acc[MAX_PKT_BWD * 8] = {0.f};
+  float db_acc[MAX_PKT_BWD * 8] = {0.f};
+  const bool cached = pkts <= MAX_PKT_BWD && D == pkts * WAVE * 8;
   for (int row = blockIdx.x * WAVES_PER_BLOCK + wid; row < N;
        row += gridDim.x * WAVES_PER_BLOCK) {
     const short* dyr = dy + (long)row * D;
     const short* xr = x + (long)row * D;
     short* dxr = dx + (long)row * D;
     const float mean = mean_in[row], rstd = rstd_in[row];
-    // bwd caches 3 f32 arrays per lane -> cap at 4 packets (D <= 2048)
-#define MAX_PKT_BWD 4
     if (pkts <= MAX_PKT_BWD && D == pkts * WAVE * 8) {
       float xh[MAX_PKT_BWD * 8], dyg[MAX_PKT_BWD * 8], dyv[MAX_PKT_BWD * 8];
       float s1 = 0.f, s2 = 0.f;
@@ -163,10 +170,8 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
         for (int j = 0; j < 8; ++j) {
           int k = p * 8 + j;
           o[j] = f32_to_bf16(rstd * (dyg[k] - s1 - xh[k] * s2));
-          // LDS partials: each (lane, j) owns column base+j exclusively within
-          // this wave; different waves share columns -> atomic LDS add.
-          atomicAdd(&sg[base + j], dyv[k] * xh[k]);
-          atomicAdd(&sb[base + j], dyv[k]);
+          dg_acc[k] += dyv[k] * xh[k];
+          db_acc[k] += dyv[k];
         }
         *(short8_t*)(dxr + base) = o;
       }
@@ -201,6 +206,18 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       }
     }
   }
+  if (cached) {
+#pragma unroll
+    for (int p = 0; p < MAX_PKT_BWD; ++p) {
+      if (p >= pkts) break;
+      int base = (p * WAVE + lane) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(&sg[base + j], dg_acc[p * 8 + j]);
+        atomicAdd(&sb[base + j], db_acc[p * 8 + j]);
+      }
+    }
+  }
   __syncthreads();
   float* og = ws_dgamma + (long)blockIdx.x * D;
   float* ob = ws_dbeta + (long)blockIdx.x * D;
@@ -208,13 +225,23 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
 }
 
 // Column-sum of the [R, D] f32 partials workspace into [D] f32.
+// Two-stage split-row reduction: the single-threaded-column version ran 4
+// blocks at D=1024 (15.6 ms/step in the baseline profile); this one fills the
+// chip with gridDim.y row-splits x float4 columns and stays deterministic.
 __global__ void __launch_bounds__(256)
-colsum_kernel(const float* __restrict__ ws, float* __restrict__ out, int R, int D) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
+colsum_stage_kernel(const float* __restrict__ ws, float* __restrict__ out,
+                    int R, int D, int rows_per_split) {
+  int col = (blockIdx.x * 256 + threadIdx.x) * 4;
   if (col >= D) return;
-  float s = 0.f;
-  for (int r = 0; r < R; ++r) s += ws[(long)r * D + col];
-  out[col] = s;
+  int r0 = blockIdx.y * rows_per_split;
+  int r1 = min(R, r0 + rows_per_split);
+  float4_t s = {0.f, 0.f, 0.f, 0.f};
+  for (int r = r0; r < r1; ++r) {
+    float4_t v = *(const float4_t*)(ws + (long)r * D + col);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) s[j] += v[j];
+  }
+  *(float4_t*)(out + (long)blockIdx.y * D + col) = s;
 }
 
 hipError_t ln_fwd_launch(const void* x, const void* gamma, const void* beta,
@@ -238,10 +265,23 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
   return hipGetLastError();
 }
 
-hipError_t colsum_launch(const void* ws, void* out, int R, int D,
-                         hipStream_t stream) {
-  int grid = (D + 255) / 256;
- hipLaunchKernelGGL(( colsum_kernel), dim3(grid), dim3(256), 0, stream, (const float*)ws, (float*)out, R, D);
+#define COLSUM_SPLITS 16
+hipError_t colsum_launch(const void* ws, void* scratch, void* out, int R,
+                         int D, hipStream_t stream) {
+  dim3 grid1((D / 4 + 255) / 256, 1);
+  if (R <= COLSUM_SPLITS) {
+   hipLaunchKernelGGL(( colsum_stage_kernel), dim3(grid1), dim3(256), 0, stream, 
+        (const float*)ws, (float*)out, R, D, R);
+    return hipGetLastError();
+  }
+  int rows_per_split = (R + COLSUM_SPLITS - 1) / COLSUM_SPLITS;
+  dim3 grid2((D / 4 + 255) / 256, COLSUM_SPLITS);
+ hipLaunchKernelGGL(( colsum_stage_kernel), dim3(grid2), dim3(256), 0, stream, 
+      (const float*)ws, (float*)scratch, R, D, rows_per_split);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return e;
+ hipLaunchKernelGGL(( colsum_stage_kernel), dim3(grid1), dim3(256), 0, stream, 
+      (const float*)scratch, (float*)out, COLSUM_SPLITS, D, COLSUM_SPLITS);
   return hipGetLastError();
 }
 

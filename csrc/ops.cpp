@@ -7,6 +7,8 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+#include <algorithm>
+#include <numeric>
 
 extern "C" {
 hipError_t ln_fwd_launch(const void*, const void*, const void*, void*, void*,
@@ -14,7 +16,7 @@ hipError_t ln_fwd_launch(const void*, const void*, const void*, void*, void*,
 hipError_t ln_bwd_launch(const void*, const void*, const void*, const void*,
                          const void*, void*, void*, void*, int, int, int,
                          hipStream_t);
-hipError_t colsum_launch(const void*, void*, int, int, hipStream_t);
+hipError_t colsum_launch(const void*, void*, void*, int, int, hipStream_t);
 hipError_t bias_gelu_fwd_launch(const void*, const void*, void*, long, int,
                                 int, hipStream_t);
 hipError_t bias_gelu_bwd_launch(const void*, const void*, const void*, void*,
@@ -88,10 +90,11 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                           cur_stream()));
   auto dgamma = torch::empty({D}, opts);
   auto dbeta = torch::empty({D}, opts);
-  CHECK_HIP(colsum_launch(ws_dg.data_ptr(), dgamma.data_ptr(), grid, D,
-                          cur_stream()));
-  CHECK_HIP(colsum_launch(ws_db.data_ptr(), dbeta.data_ptr(), grid, D,
-                          cur_stream()));
+  auto scratch = torch::empty({16, D}, opts);
+  CHECK_HIP(colsum_launch(ws_dg.data_ptr(), scratch.data_ptr(),
+                          dgamma.data_ptr(), grid, D, cur_stream()));
+  CHECK_HIP(colsum_launch(ws_db.data_ptr(), scratch.data_ptr(),
+                          dbeta.data_ptr(), grid, D, cur_stream()));
   return {dx, dgamma, dbeta};
 }
 
@@ -114,13 +117,19 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const long n = x.numel();
   auto dx = torch::empty_like(x);
   int grid = (int)std::min<long>((n / 8 + 255) / 256, 1024);
+  // round the grid up to a multiple of D/gcd(D, 2048) so the kernel's grid
+  // stride is a multiple of D -> fixed per-thread column window (fast path)
+  long g = std::__gcd((long)D, 2048L);
+  long g0 = D / g;
+  if (g0 <= 2048) grid = (int)((grid + g0 - 1) / g0 * g0);
   auto ws = torch::empty({grid, D}, x.options().dtype(torch::kFloat32));
   CHECK_HIP(bias_gelu_bwd_launch(dy.data_ptr(), x.data_ptr(), b.data_ptr(),
                                  dx.data_ptr(), ws.data_ptr(), n, D, grid,
                                  cur_stream()));
   auto dbias = torch::empty({D}, x.options().dtype(torch::kFloat32));
-  CHECK_HIP(colsum_launch(ws.data_ptr(), dbias.data_ptr(), grid, D,
-                          cur_stream()));
+  auto scratch = torch::empty({16, D}, x.options().dtype(torch::kFloat32));
+  CHECK_HIP(colsum_launch(ws.data_ptr(), scratch.data_ptr(), dbias.data_ptr(),
+                          grid, D, cur_stream()));
   return {dx, dbias};
 }
 

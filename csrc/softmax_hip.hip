@@ -38,10 +38,15 @@ softmax_fwd_kernel(const short* __restrict__ s_in, const float* __restrict__ mas
         if (p >= pkts) break;
         int base = (p * WAVE + lane) * 8;
         short8_t x = *(const short8_t*)(sr + base);
+        float4_t m0, m1;
+        if (mrow) {
+          m0 = *(const float4_t*)(mrow + base);
+          m1 = *(const float4_t*)(mrow + base + 4);
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float f = bf16_to_f32(x[j]) * scale;
-          if (mrow) f += mrow[base + j];
+          if (mrow) f += j < 4 ? m0[j] : m1[j - 4];
           v[p * 8 + j] = f;
           m = fmaxf(m, f);
         }

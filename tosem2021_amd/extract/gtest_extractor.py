@@ -15,8 +15,12 @@ from tosem2021_amd.extract.python_extractor import Assertion, TestCase
 RE_TEST_MACRO = re.compile(
     r"\b(TEST|TEST_F|TEST_P|TYPED_TEST|TYPED_TEST_P|INSTANTIATE_TEST_SUITE_P)"
     r"\s*\(\s*([A-Za-z_]\w*)\s*,\s*([A-Za-z_]\w*)\s*\)")
+# boost.test (kenlm, openfst in the DeepSpeech snapshot): one-arg case macro
+RE_BOOST_CASE = re.compile(
+    r"\b(BOOST_AUTO_TEST_CASE|BOOST_FIXTURE_TEST_CASE|BOOST_AUTO_TEST_CASE_TEMPLATE)"
+    r"\s*\(\s*([A-Za-z_]\w*)")
 RE_ASSERT = re.compile(
-    r"\b((?:EXPECT|ASSERT)_[A-Z_0-9]+)\s*\(")
+    r"\b((?:EXPECT|ASSERT)_[A-Z_0-9]+|BOOST_(?:CHECK|REQUIRE|WARN)(?:_[A-Z_0-9]+)?)\s*\(")
 RE_DEATH = re.compile(r"DEATH|THROW", re.I)
 
 
@@ -85,10 +89,13 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     except OSError:
         return []
     cases: List[TestCase] = []
+    marks = []
     for m in RE_TEST_MACRO.finditer(text):
-        macro, suite, name = m.group(1), m.group(2), m.group(3)
-        if macro == "INSTANTIATE_TEST_SUITE_P":
-            continue
+        if m.group(1) != "INSTANTIATE_TEST_SUITE_P":
+            marks.append((m, m.group(2), m.group(3)))
+    for m in RE_BOOST_CASE.finditer(text):
+        marks.append((m, "", m.group(2)))
+    for m, suite, name in sorted(marks, key=lambda t: t[0].start()):
         brace = text.find("{", m.end())
         if brace < 0:
             continue
@@ -106,7 +113,8 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
                 inner = src[src.find("(") + 1:]
                 parts = inner.rsplit(",", 1)
                 exc = parts[-1].strip(" );") if len(parts) > 1 else ""
-            if "NEAR" in call or "FLOAT_EQ" in call or "DOUBLE_EQ" in call:
+            if "NEAR" in call or "FLOAT_EQ" in call or "DOUBLE_EQ" in call \
+                or "CLOSE" in call:
                 kind = "approx"
             assertions.append(Assertion(
                 kind=kind, call_name=call, source=src[:500], lineno=a_line,

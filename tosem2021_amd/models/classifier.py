@@ -95,15 +95,16 @@ class Attention(nn.Module):
 
     def forward(self, x, mask_bias: Optional[torch.Tensor]):
         B, L, D = x.shape
-        qkv = self.qkv(x).view(B, L, 3, self.n_heads, self.head_dim)
-        q, k, v = qkv.unbind(dim=2)                       # [B, L, H, dh]
-        q = q.permute(0, 2, 1, 3)                         # [B, H, L, dh]
-        k = k.permute(0, 2, 3, 1)                         # [B, H, dh, L]
-        v = v.permute(0, 2, 1, 3)
-        scores = torch.matmul(q, k)                       # [B, H, L, L]
+        # one contiguous repack [3, B, H, L, dh] instead of three strided bmm
+        # operands (strided 4D matmul was materializing batched copies every
+        # layer — profiles/r01_kernel_stats_baseline.md, CatArrayBatchedCopy)
+        qkv = (self.qkv(x).view(B, L, 3, self.n_heads, self.head_dim)
+               .permute(2, 0, 3, 1, 4).contiguous())
+        q, k, v = qkv.unbind(0)                           # [B, H, L, dh]
+        scores = torch.matmul(q, k.transpose(-1, -2))     # [B, H, L, L]
         p = ops.fused_softmax(scores, mask_bias, self.scale)
         out = torch.matmul(p, v)                          # [B, H, L, dh]
-        out = out.permute(0, 2, 1, 3).reshape(B, L, D)
+        out = out.transpose(1, 2).contiguous().view(B, L, D)
         return self.proj(out)
 
 

@@ -12,6 +12,8 @@ from tosem2021_amd.corpus import walker
 from tosem2021_amd.corpus.registry import PROJECTS, Project, project_root
 from tosem2021_amd.extract.python_extractor import extract_file
 from tosem2021_amd.extract.schema import TAXONOMY_COLUMNS, TestCaseRow
+from tosem2021_amd.utils.metrics import get_metrics
+from tosem2021_amd.utils.trace import trace
 
 
 def mine_project(key: str, corpus_root: Optional[str] = None,
@@ -23,22 +25,24 @@ def mine_project(key: str, corpus_root: Optional[str] = None,
     if not os.path.isdir(root):
         raise FileNotFoundError(f"project snapshot not found: {root}")
     rows: List[TestCaseRow] = []
-    files = walker.test_files(root, languages)
-    if max_files:
-        files = files[:max_files]
-    for file_id, f in enumerate(files):
-        if f.language == "python":
-            cases = extract_file(f.path, f.rel)
-        elif f.language == "cpp":
-            from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
-            cases = extract_gtest_file(f.path, f.rel)
-        elif f.language == "ts":
-            from tosem2021_amd.extract.ts_extractor import extract_ts_file
-            cases = extract_ts_file(f.path, f.rel)
-        else:
-            continue
-        for case in cases:
-            rows.extend(classify_case(case, repo=proj.name, file_id=file_id))
+    with trace("mine_project", project=key):
+        files = walker.test_files(root, languages)
+        if max_files:
+            files = files[:max_files]
+        for file_id, f in enumerate(files):
+            if f.language == "python":
+                cases = extract_file(f.path, f.rel)
+            elif f.language == "cpp":
+                from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+                cases = extract_gtest_file(f.path, f.rel)
+            elif f.language == "ts":
+                from tosem2021_amd.extract.ts_extractor import extract_ts_file
+                cases = extract_ts_file(f.path, f.rel)
+            else:
+                continue
+            for case in cases:
+                rows.extend(classify_case(case, repo=proj.name, file_id=file_id))
+    get_metrics().inc("rows_mined", len(rows), repo=proj.name)
     for i, r in enumerate(rows):
         r.index = i + 1
     return rows

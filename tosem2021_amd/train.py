@@ -25,6 +25,8 @@ import torch.distributed as dist
 from tosem2021_amd import ops
 from tosem2021_amd.models import MLTC, MLTCConfig, build_model
 from tosem2021_amd.parallel.ddp import BucketedAllReduce
+from tosem2021_amd.utils.metrics import get_metrics
+from tosem2021_amd.utils.trace import trace
 
 PAD_ELEMS = 512  # flat buffers padded so the fused AdamW's 4-wide loop is exact
 
@@ -109,17 +111,21 @@ class Trainer:
     # ---- one optimization step ----------------------------------------------
     def step(self, tokens: torch.Tensor, attn_mask: Optional[torch.Tensor],
              labels: Dict[str, torch.Tensor]) -> float:
-        self.flat.zero_grad()
-        logits = self.model(tokens, attn_mask)
-        loss = self.model.loss(logits, labels)
-        loss.backward()
-        self.ddp.finalize()
-        self.step_num += 1
-        ops.adamw_step(
-            self.flat.flat, self.flat.grad_flat, self.flat.m, self.flat.v,
-            self.flat.master, lr=self._lr(), beta1=self.cfg.beta1,
-            beta2=self.cfg.beta2, eps=self.cfg.eps, wd=self.cfg.weight_decay,
-            step=self.step_num, grad_scale=self.ddp.grad_scale)
+        with trace("train_step", step=self.step_num + 1):
+            self.flat.zero_grad()
+            logits = self.model(tokens, attn_mask)
+            loss = self.model.loss(logits, labels)
+            loss.backward()
+            self.ddp.finalize()
+            self.step_num += 1
+            lr = self._lr()
+            ops.adamw_step(
+                self.flat.flat, self.flat.grad_flat, self.flat.m, self.flat.v,
+                self.flat.master, lr=lr, beta1=self.cfg.beta1,
+                beta2=self.cfg.beta2, eps=self.cfg.eps, wd=self.cfg.weight_decay,
+                step=self.step_num, grad_scale=self.ddp.grad_scale)
+        get_metrics().observe_step(self.step_num, loss=float(loss.detach()),
+                                   lr=lr)
         if self.cfg.ckpt_every and self.cfg.ckpt_dir and \
                 self.step_num % self.cfg.ckpt_every == 0:
             self.save()
