@@ -1,0 +1,62 @@
+"""Selection funnel: unit tests + golden shape checks on the reference data."""
+import os
+
+import pandas as pd
+import pytest
+
+from tosem2021_amd.corpus.selection import (
+    DEFAULT_ROUNDS, FunnelCriteria, apply_criteria, load_metrics, run_funnel)
+
+
+def _toy():
+    return pd.DataFrame({
+        "Repos": ["a/x", "b/y", "c/z"],
+        "topic": ["ml", "ml", "ml"],
+        "commits": [5000, 100, 2000],
+        "contributors": [50, 2, 30],
+        "issues": [10, 1, 5],
+        "pulls": [10, 1, 5],
+        "releases": [5, 0, 2],
+        "size": [1, 1, 1],
+        "stars": [2000, 10, 1500],
+        "forks": [10, 1, 5],
+        "open_issues": [1, 0, 2],
+        "archived": [False, False, True],
+        "created_at": ["2017"] * 3,
+        "updated_at": ["2021"] * 3,
+        "language": ["Python", "Python", "C++"],
+    })
+
+
+def test_apply_criteria():
+    df = _toy()
+    out = apply_criteria(df, FunnelCriteria(min_stars=1000, min_commits=1000,
+                                            min_contributors=20,
+                                            min_releases=1))
+    assert list(out["Repos"]) == ["a/x"]  # c/z archived, b/y inactive
+    out = apply_criteria(df, FunnelCriteria(exclude_archived=False,
+                                            languages=["C++"]))
+    assert list(out["Repos"]) == ["c/z"]
+
+
+def test_run_funnel_monotone():
+    df = _toy()
+    rounds = [FunnelCriteria(), FunnelCriteria(min_stars=1000)]
+    outs = run_funnel(df, rounds)
+    assert len(outs) == 2
+    assert len(outs[0]) >= len(outs[1])
+
+
+def test_reference_funnel_tables(reference_root):
+    base = os.path.join(reference_root, "selection", "Reposition")
+    v3 = load_metrics(os.path.join(base, "Repos_metrics_v3.csv"))
+    assert len(v3) == 311  # 312 lines incl. header
+    outs = run_funnel(v3, DEFAULT_ROUNDS)
+    # the funnel must be strictly narrowing on the real table and keep the
+    # nine studied systems' orgs in the surviving set
+    assert len(outs[0]) <= len(v3)
+    assert len(outs[-1]) < len(outs[0])
+    survivors = set(outs[-1]["Repos"])
+    for org in ("ApolloAuto/apollo", "ray-project/ray", "microsoft/nni"):
+        if org in set(v3["Repos"]):
+            assert org in survivors, org
