@@ -20,10 +20,11 @@ bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
   for (long i = idx0; i < n_elem; i += stride) {
     short8_t v = *(const short8_t*)(x + i);
     int col = (int)(i % D);  // D % 8 == 0 so the packet stays in one row
+    short8_t b8 = *(const short8_t*)(b + col);
     short8_t o;
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[j]) + bf16_to_f32(b[col + j])));
+      o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[j]) + bf16_to_f32(b8[j])));
     *(short8_t*)(y + i) = o;
   }
 }
@@ -45,11 +46,10 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   if (stride % D == 0) {
     float acc[8] = {0.f};
     const int col = (int)(idx0 % D);
-    const float b0 = bf16_to_f32(b[col + 0]), b1 = bf16_to_f32(b[col + 1]),
-                b2 = bf16_to_f32(b[col + 2]), b3 = bf16_to_f32(b[col + 3]),
-                b4 = bf16_to_f32(b[col + 4]), b5 = bf16_to_f32(b[col + 5]),
-                b6 = bf16_to_f32(b[col + 6]), b7 = bf16_to_f32(b[col + 7]);
-    const float bb[8] = {b0, b1, b2, b3, b4, b5, b6, b7};
+    short8_t bv = *(const short8_t*)(b + col);
+    float bb[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) bb[j] = bf16_to_f32(bv[j]);
     for (long i = idx0; i < n_elem; i += stride) {
       short8_t vd = *(const short8_t*)(dy + i);
       short8_t vx = *(const short8_t*)(x + i);
@@ -70,11 +70,12 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       short8_t vd = *(const short8_t*)(dy + i);
       short8_t vx = *(const short8_t*)(x + i);
       int col = (int)(i % D);
+      short8_t b8 = *(const short8_t*)(b + col);
       short8_t o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = bf16_to_f32(vd[j]);
-        float pre = bf16_to_f32(vx[j]) + bf16_to_f32(b[col + j]);
+        float pre = bf16_to_f32(vx[j]) + bf16_to_f32(b8[j]);
         o[j] = f32_to_bf16(g * gelu_tanh_grad(pre));
         atomicAdd(&sdb[col + j], g);
       }

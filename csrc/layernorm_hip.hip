@@ -23,8 +23,10 @@
 extern "C" {
 
 __global__ void __launch_bounds__(BLOCK)
-ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ gamma,
+ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
+              const short* __restrict__ gamma,
               const short* __restrict__ beta, short* __restrict__ y,
+              short* __restrict__ s_out,
               float* __restrict__ mean_out, float* __restrict__ rstd_out,
               int N, int D, float eps) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -33,20 +35,37 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ gamma,
   for (int row = blockIdx.x * WAVES_PER_BLOCK + wid; row < N;
        row += gridDim.x * WAVES_PER_BLOCK) {
     const short* xr = x + (long)row * D;
+    const short* rr = res ? res + (long)row * D : nullptr;
     short* yr = y + (long)row * D;
+    short* sr = s_out ? s_out + (long)row * D : nullptr;
     float vals[MAX_PKT * 8];
     float s = 0.f;
     if (pkts <= MAX_PKT && D == pkts * WAVE * 8) {
 #pragma unroll
       for (int p = 0; p < MAX_PKT; ++p) {
         if (p >= pkts) break;
-        short8_t v = *(const short8_t*)(xr + (p * WAVE + lane) * 8);
+        int base = (p * WAVE + lane) * 8;
+        short8_t v = *(const short8_t*)(xr + base);
+        short8_t so;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float f = bf16_to_f32(v[j]);
           vals[p * 8 + j] = f;
-          s += f;
         }
+        if (rr) {
+          short8_t rv = *(const short8_t*)(rr + base);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            // bf16-rounded sum so s_out, the saved x for bwd, and the stats
+            // all see the SAME residual-stream value
+            short sum_b = f32_to_bf16(vals[p * 8 + j] + bf16_to_f32(rv[j]));
+            vals[p * 8 + j] = bf16_to_f32(sum_b);
+            so[j] = sum_b;
+          }
+          *(short8_t*)(sr + base) = so;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s += vals[p * 8 + j];
       }
       float mean = wave_sum(s) / (float)D;
       float var = 0.f;
@@ -66,11 +85,13 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ gamma,
       for (int p = 0; p < MAX_PKT; ++p) {
         if (p >= pkts) break;
         int base = (p * WAVE + lane) * 8;
+        short8_t g8 = *(const short8_t*)(gamma + base);
+        short8_t b8 = *(const short8_t*)(beta + base);
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float xhat = (vals[p * 8 + j] - mean) * rstd;
-          o[j] = f32_to_bf16(xhat * bf16_to_f32(gamma[base + j]) + bf16_to_f32(beta[base + j]));
+          o[j] = f32_to_bf16(xhat * bf16_to_f32(g8[j]) + bf16_to_f32(b8[j]));
         }
         *(short8_t*)(yr + base) = o;
       }
@@ -78,13 +99,23 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ gamma,
       // general path: two passes over the row, 8-wide loads, any D % 8 == 0
       for (int i = lane * 8; i < D; i += WAVE * 8) {
         short8_t v = *(const short8_t*)(xr + i);
+        if (rr) {
+          short8_t rv = *(const short8_t*)(rr + i);
+          short8_t so;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            so[j] = f32_to_bf16(bf16_to_f32(v[j]) + bf16_to_f32(rv[j]));
+          }
+          *(short8_t*)(sr + i) = so;
+          v = so;
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) s += bf16_to_f32(v[j]);
       }
       float mean = wave_sum(s) / (float)D;
       float var = 0.f;
       for (int i = lane * 8; i < D; i += WAVE * 8) {
-        short8_t v = *(const short8_t*)(xr + i);
+        short8_t v = *(const short8_t*)((rr ? sr : xr) + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float d = bf16_to_f32(v[j]) - mean;
@@ -95,12 +126,14 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ gamma,
       float rstd = rsqrtf(var + eps);
       if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
       for (int i = lane * 8; i < D; i += WAVE * 8) {
-        short8_t v = *(const short8_t*)(xr + i);
+        short8_t v = *(const short8_t*)((rr ? sr : xr) + i);
+        short8_t g8 = *(const short8_t*)(gamma + i);
+        short8_t b8 = *(const short8_t*)(beta + i);
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float xhat = (bf16_to_f32(v[j]) - mean) * rstd;
-          o[j] = f32_to_bf16(xhat * bf16_to_f32(gamma[i + j]) + bf16_to_f32(beta[i + j]));
+          o[j] = f32_to_bf16(xhat * bf16_to_f32(g8[j]) + bf16_to_f32(b8[j]));
         }
         *(short8_t*)(yr + i) = o;
       }
@@ -150,11 +183,12 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
         int base = (p * WAVE + lane) * 8;
         short8_t vd = *(const short8_t*)(dyr + base);
         short8_t vx = *(const short8_t*)(xr + base);
+        short8_t g8 = *(const short8_t*)(gamma + base);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float d = bf16_to_f32(vd[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
-          float g = d * bf16_to_f32(gamma[base + j]);
+          float g = d * bf16_to_f32(g8[j]);
           xh[p * 8 + j] = h; dyg[p * 8 + j] = g; dyv[p * 8 + j] = d;
           s1 += g; s2 += g * h;
         }
@@ -180,9 +214,10 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       for (int i = lane * 8; i < D; i += WAVE * 8) {
         short8_t vd = *(const short8_t*)(dyr + i);
         short8_t vx = *(const short8_t*)(xr + i);
+        short8_t g8 = *(const short8_t*)(gamma + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float g = bf16_to_f32(vd[j]) * bf16_to_f32(gamma[i + j]);
+          float g = bf16_to_f32(vd[j]) * bf16_to_f32(g8[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
           s1 += g; s2 += g * h;
         }
@@ -192,12 +227,13 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       for (int i = lane * 8; i < D; i += WAVE * 8) {
         short8_t vd = *(const short8_t*)(dyr + i);
         short8_t vx = *(const short8_t*)(xr + i);
+        short8_t g8 = *(const short8_t*)(gamma + i);
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float d = bf16_to_f32(vd[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
-          float g = d * bf16_to_f32(gamma[i + j]);
+          float g = d * bf16_to_f32(g8[j]);
           o[j] = f32_to_bf16(rstd * (g - s1 - h * s2));
           atomicAdd(&sg[i + j], d * h);
           atomicAdd(&sb[i + j], d);
@@ -244,12 +280,14 @@ colsum_stage_kernel(const float* __restrict__ ws, float* __restrict__ out,
   *(float4_t*)(out + (long)blockIdx.y * D + col) = s;
 }
 
-hipError_t ln_fwd_launch(const void* x, const void* gamma, const void* beta,
-                         void* y, void* mean, void* rstd, int N, int D,
-                         float eps, int grid, hipStream_t stream) {
- hipLaunchKernelGGL(( ln_fwd_kernel), dim3(grid), dim3(BLOCK), 0, stream, (const short*)x, (const short*)gamma,
-                                            (const short*)beta, (short*)y,
-                                            (float*)mean, (float*)rstd, N, D, eps);
+hipError_t ln_fwd_launch(const void* x, const void* res, const void* gamma,
+                         const void* beta, void* y, void* s_out, void* mean,
+                         void* rstd, int N, int D, float eps, int grid,
+                         hipStream_t stream) {
+ hipLaunchKernelGGL(( ln_fwd_kernel), dim3(grid), dim3(BLOCK), 0, stream, 
+      (const short*)x, (const short*)res, (const short*)gamma,
+      (const short*)beta, (short*)y, (short*)s_out, (float*)mean,
+      (float*)rstd, N, D, eps);
   return hipGetLastError();
 }
 

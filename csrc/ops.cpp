@@ -11,8 +11,9 @@
 #include <numeric>
 
 extern "C" {
-hipError_t ln_fwd_launch(const void*, const void*, const void*, void*, void*,
-                         void*, int, int, float, int, hipStream_t);
+hipError_t ln_fwd_launch(const void*, const void*, const void*, const void*,
+                         void*, void*, void*, void*, int, int, float, int,
+                         hipStream_t);
 hipError_t ln_bwd_launch(const void*, const void*, const void*, const void*,
                          const void*, void*, void*, void*, int, int, int,
                          hipStream_t);
@@ -56,7 +57,9 @@ void check_f32(const torch::Tensor& t, const char* name) {
 
 }  // namespace
 
-std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x,
+                                         c10::optional<torch::Tensor> residual,
+                                         torch::Tensor gamma,
                                          torch::Tensor beta, double eps) {
   check_bf16(x, "x"); check_bf16(gamma, "gamma"); check_bf16(beta, "beta");
   const int D = (int)x.size(-1);
@@ -67,10 +70,23 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({N}, opts);
   auto rstd = torch::empty({N}, opts);
   int grid = (int)std::min<long>((N + 3) / 4, 1024);
-  CHECK_HIP(ln_fwd_launch(x.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
-                          y.data_ptr(), mean.data_ptr(), rstd.data_ptr(),
-                          (int)N, D, (float)eps, grid, cur_stream()));
-  return {y, mean, rstd};
+  const void* res_ptr = nullptr;
+  torch::Tensor s_out;
+  void* s_ptr = nullptr;
+  if (residual.has_value()) {
+    check_bf16(*residual, "residual");
+    TORCH_CHECK(residual->sizes() == x.sizes(), "residual shape mismatch");
+    res_ptr = residual->data_ptr();
+    s_out = torch::empty_like(x);
+    s_ptr = s_out.data_ptr();
+  } else {
+    s_out = x;  // residual stream unchanged
+  }
+  CHECK_HIP(ln_fwd_launch(x.data_ptr(), res_ptr, gamma.data_ptr(),
+                          beta.data_ptr(), y.data_ptr(), s_ptr,
+                          mean.data_ptr(), rstd.data_ptr(), (int)N, D,
+                          (float)eps, grid, cur_stream()));
+  return {y, s_out, mean, rstd};
 }
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,

@@ -1,4 +1,5 @@
 """Diagnose bias_gelu numerics vs the fp32 reference (run on GPU box)."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 from tosem2021_amd import ops

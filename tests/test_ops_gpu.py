@@ -136,3 +136,21 @@ def test_grad_scale_matches_prescaled():
     ops.hip_ops().adamw_step(pb, _bf16(g * 0.25), sb, vb, mb, 1e-2, 0.9,
                              0.999, 1e-8, 0.0, 1, 1.0)
     assert torch.allclose(ma, mb, atol=2e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("N,D", [(64, 1024), (33, 136)])
+def test_add_layernorm_fused(N, D):
+    torch.manual_seed(7)
+    xg = _bf16(torch.randn(N, D))
+    rg = _bf16(torch.randn(N, D))
+    gg = _bf16(torch.randn(D))
+    bg = _bf16(torch.randn(D))
+    y, s, mean, rstd = ops.hip_ops().layernorm_fwd(xg, rg, gg, bg, 1e-5)
+    ye, se, me, re_ = ref.add_layernorm_fwd(xg.cpu(), rg.cpu(), gg.cpu(),
+                                            bg.cpu(), 1e-5)
+    assert torch.equal(s.cpu(), se), "fused residual sum must match bf16 add"
+    assert torch.allclose(y.float().cpu(), ye.float(), atol=3e-2, rtol=2e-2)
+    assert torch.allclose(mean.cpu(), me, atol=2e-3, rtol=1e-3)
+    # no-residual call returns x itself as the stream
+    y2, s2, _, _ = ops.hip_ops().layernorm_fwd(xg, None, gg, bg, 1e-5)
+    assert s2.data_ptr() == xg.data_ptr()

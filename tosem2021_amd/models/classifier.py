@@ -124,6 +124,14 @@ class FFN(nn.Module):
 
 
 class EncoderBlock(nn.Module):
+    """Pre-LN block carrying an un-added (stream, delta) residual pair.
+
+    The residual add is fused into the next LayerNorm's first read
+    (ops.fused_add_layernorm), so the block never launches a separate
+    elementwise add: it receives the previous sublayer's (x, delta), and
+    returns its own.
+    """
+
     def __init__(self, cfg: MLTCConfig):
         super().__init__()
         self.ln1 = FusedLayerNorm(cfg.d_model, cfg.layer_norm_eps)
@@ -132,15 +140,22 @@ class EncoderBlock(nn.Module):
         self.ffn = FFN(cfg)
         self.dropout = cfg.dropout
 
-    def forward(self, x, mask_bias):
-        h = self.attn(self.ln1(x), mask_bias)
-        if self.dropout:
-            h = F.dropout(h, self.dropout, self.training)
-        x = x + h
-        h = self.ffn(self.ln2(x))
-        if self.dropout:
-            h = F.dropout(h, self.dropout, self.training)
-        return x + h
+    def _drop(self, h):
+        return F.dropout(h, self.dropout, self.training) if self.dropout else h
+
+    def forward(self, x, delta, mask_bias):
+        if delta is None:
+            y1 = ops.fused_layernorm(x, self.ln1.weight, self.ln1.bias,
+                                     self.ln1.eps)
+            s1 = x
+        else:
+            y1, s1 = ops.fused_add_layernorm(x, delta, self.ln1.weight,
+                                             self.ln1.bias, self.ln1.eps)
+        h = self._drop(self.attn(y1, mask_bias))
+        y2, s2 = ops.fused_add_layernorm(s1, h, self.ln2.weight,
+                                         self.ln2.bias, self.ln2.eps)
+        h2 = self._drop(self.ffn(y2))
+        return s2, h2
 
 
 class MLTC(nn.Module):
@@ -174,9 +189,14 @@ class MLTC(nn.Module):
             mask_bias = torch.where(
                 attn_mask, 0.0, -1e9
             ).to(torch.float32).contiguous()
+        delta = None
         for blk in self.blocks:
-            x = blk(x, mask_bias)
-        x = self.ln_f(x)
+            x, delta = blk(x, delta, mask_bias)
+        if delta is None:
+            x = self.ln_f(x)
+        else:
+            x, _ = ops.fused_add_layernorm(x, delta, self.ln_f.weight,
+                                           self.ln_f.bias, self.ln_f.eps)
         # masked mean-pool over valid tokens
         if attn_mask is not None:
             w = attn_mask.to(x.dtype).unsqueeze(-1)

@@ -48,7 +48,7 @@ class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, eps):
         if x.is_cuda:
-            y, mean, rstd = hip_ops().layernorm_fwd(x, gamma, beta, eps)
+            y, _, mean, rstd = hip_ops().layernorm_fwd(x, None, gamma, beta, eps)
         else:
             y, mean, rstd = reference.layernorm_fwd(x, gamma, beta, eps)
         ctx.save_for_backward(x, gamma, mean, rstd)
@@ -67,6 +67,40 @@ class _LayerNormFn(torch.autograd.Function):
 
 def fused_layernorm(x, gamma, beta, eps: float = 1e-5):
     return _LayerNormFn.apply(x.contiguous(), gamma, beta, eps)
+
+
+class _AddLayerNormFn(torch.autograd.Function):
+    """(y, s) = (LN(x + residual), x + residual) — the residual-stream add is
+    fused into the LN kernel's first read (one kernel, no separate add)."""
+
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, eps):
+        if x.is_cuda:
+            y, s, mean, rstd = hip_ops().layernorm_fwd(x, residual, gamma,
+                                                       beta, eps)
+        else:
+            y, s, mean, rstd = reference.add_layernorm_fwd(x, residual, gamma,
+                                                           beta, eps)
+        ctx.save_for_backward(s, gamma, mean, rstd)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, gamma, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if s.is_cuda:
+            dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy, s, gamma, mean, rstd)
+        else:
+            dx, dgamma, dbeta = reference.layernorm_bwd(dy, s, gamma, mean, rstd)
+        if ds is not None:
+            dx = dx + ds
+        return dx, dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
+
+
+def fused_add_layernorm(x, residual, gamma, beta, eps: float = 1e-5):
+    """Returns (y, s): y = LN(x+residual), s = the new residual stream."""
+    return _AddLayerNormFn.apply(x.contiguous(), residual.contiguous(),
+                                 gamma, beta, eps)
 
 
 class _BiasGeluFn(torch.autograd.Function):

@@ -25,6 +25,13 @@ def layernorm_fwd(
     return y.to(x.dtype), mean.reshape(-1), rstd.reshape(-1)
 
 
+def add_layernorm_fwd(x, residual, gamma, beta, eps):
+    """s = x + residual (rounded to x.dtype); y = LN(s). Returns (y, s, mean, rstd)."""
+    s = (x.float() + residual.float()).to(x.dtype)
+    y, mean, rstd = layernorm_fwd(s, gamma, beta, eps)
+    return y, s, mean, rstd
+
+
 def layernorm_bwd(
     dy: torch.Tensor,
     x: torch.Tensor,
