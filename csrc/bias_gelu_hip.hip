@@ -59,8 +59,9 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       for (int j = 0; j < 8; ++j) {
         float g = bf16_to_f32(vd[j]);
         float pre = bf16_to_f32(vx[j]) + bb[j];
-        o[j] = f32_to_bf16(g * gelu_tanh_grad(pre));
-        acc[j] += g;
+        float dpre = g * gelu_tanh_grad(pre);
+        o[j] = f32_to_bf16(dpre);
+        acc[j] += dpre;  // dL/db = sum over rows of dy * gelu'(pre)
       }
       *(short8_t*)(dx + i) = o;
     }
@@ -77,8 +78,9 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       for (int j = 0; j < 8; ++j) {
         float g = bf16_to_f32(vd[j]);
         float pre = bf16_to_f32(vx[j]) + bf16_to_f32(b8[j]);
-        o[j] = f32_to_bf16(g * gelu_tanh_grad(pre));
-        atomicAdd(&sdb[col + j], g);
+        float dpre = g * gelu_tanh_grad(pre);
+        o[j] = f32_to_bf16(dpre);
+        atomicAdd(&sdb[col + j], dpre);
       }
       *(short8_t*)(dx + i) = o;
     }

@@ -29,6 +29,12 @@ hipError_t softmax_bwd_launch(const void*, const void*, void*, long, int,
 hipError_t adamw_launch(void*, const void*, int, void*, void*, void*, long,
                         float, float, float, float, float, int, float, int,
                         hipStream_t);
+hipError_t qkv_repack_launch(const void*, void*, int, int, int, int, int,
+                             hipStream_t);
+hipError_t out_repack_launch(const void*, void*, int, int, int, int, int,
+                             hipStream_t);
+hipError_t qkv_repack_bwd3_launch(const void*, const void*, const void*,
+                                  void*, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -204,7 +210,71 @@ void adamw_step(torch::Tensor p, torch::Tensor grad, torch::Tensor m,
                          (int)step, (float)grad_scale, grid, cur_stream()));
 }
 
+torch::Tensor qkv_repack(torch::Tensor qkv, long H, bool backward) {
+  check_bf16(qkv, "qkv");
+  long B, L, dh;
+  torch::Tensor out;
+  if (!backward) {
+    // [B, L, 3*H*dh] or [B, L, 3, H, dh] -> [3, B, H, L, dh]
+    B = qkv.size(0); L = qkv.size(1);
+    dh = qkv.numel() / (B * L * 3 * H);
+    out = torch::empty({3, B, H, L, dh}, qkv.options());
+  } else {
+    // grad [3, B, H, L, dh] -> [B, L, 3*H*dh]
+    B = qkv.size(1); L = qkv.size(3);
+    dh = qkv.size(4);
+    out = torch::empty({B, L, 3 * H * dh}, qkv.options());
+  }
+  TORCH_CHECK(dh % 8 == 0, "head_dim must be a multiple of 8");
+  CHECK_HIP(qkv_repack_launch(qkv.data_ptr(), out.data_ptr(), (int)B, (int)L,
+                              (int)H, (int)dh, backward ? 1 : 0,
+                              cur_stream()));
+  return out;
+}
+
+torch::Tensor qkv_repack_bwd3(torch::Tensor dq, torch::Tensor dk,
+                              torch::Tensor dv) {
+  check_bf16(dq, "dq"); check_bf16(dk, "dk"); check_bf16(dv, "dv");
+  long B = dq.size(0), H = dq.size(1), L = dq.size(2), dh = dq.size(3);
+  auto out = torch::empty({B, L, 3 * H * dh}, dq.options());
+  CHECK_HIP(qkv_repack_bwd3_launch(dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                                   out.data_ptr(), (int)B, (int)L, (int)H,
+                                   (int)dh, cur_stream()));
+  return out;
+}
+
+torch::Tensor out_repack(torch::Tensor x, bool backward) {
+  check_bf16(x, "x");
+  long B, L, H, dh;
+  torch::Tensor out;
+  if (!backward) {
+    // [B, H, L, dh] -> [B, L, H*dh]
+    B = x.size(0); H = x.size(1); L = x.size(2); dh = x.size(3);
+    out = torch::empty({B, L, H * dh}, x.options());
+  } else {
+    TORCH_CHECK(false, "pass the 4-D grad with explicit dims via out_repack_bwd");
+  }
+  TORCH_CHECK(dh % 8 == 0, "head_dim must be a multiple of 8");
+  CHECK_HIP(out_repack_launch(x.data_ptr(), out.data_ptr(), (int)B, (int)L,
+                              (int)H, (int)dh, 0, cur_stream()));
+  return out;
+}
+
+torch::Tensor out_repack_bwd(torch::Tensor dgrad, long H) {
+  check_bf16(dgrad, "dgrad");
+  long B = dgrad.size(0), L = dgrad.size(1);
+  long dh = dgrad.numel() / (B * L * H);
+  auto out = torch::empty({B, H, L, dh}, dgrad.options());
+  CHECK_HIP(out_repack_launch(dgrad.data_ptr(), out.data_ptr(), (int)B,
+                              (int)L, (int)H, (int)dh, 1, cur_stream()));
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");
+  m.def("qkv_repack_bwd3", &qkv_repack_bwd3, "qkv repack bwd from dq,dk,dv");
+  m.def("out_repack", &out_repack, "attention output merge");
+  m.def("out_repack_bwd", &out_repack_bwd, "attention output merge bwd");
   m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (bf16, gfx950)");
   m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU fwd");

@@ -20,6 +20,7 @@ ext = CUDAExtension(
         "csrc/bias_gelu.hip",
         "csrc/softmax.hip",
         "csrc/adamw.hip",
+        "csrc/repack.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

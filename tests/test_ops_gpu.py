@@ -154,3 +154,27 @@ def test_add_layernorm_fused(N, D):
     # no-residual call returns x itself as the stream
     y2, s2, _, _ = ops.hip_ops().layernorm_fwd(xg, None, gg, bg, 1e-5)
     assert s2.data_ptr() == xg.data_ptr()
+
+
+def test_qkv_repack_roundtrip():
+    torch.manual_seed(8)
+    B, L, H, dh = 3, 40, 4, 16
+    qkv = _bf16(torch.randn(B, L, 3 * H * dh))
+    q = ops.hip_ops().qkv_repack(qkv, H, False)
+    expect = (qkv.view(B, L, 3, H, dh).permute(2, 0, 3, 1, 4).contiguous())
+    assert torch.equal(q, expect)
+    # bwd3 gathers back to the Linear layout
+    dq, dk, dv = (x.contiguous() for x in expect.unbind(0))
+    back = ops.hip_ops().qkv_repack_bwd3(dq, dk, dv)
+    assert torch.equal(back, qkv)
+
+
+def test_out_repack_roundtrip():
+    torch.manual_seed(9)
+    B, H, L, dh = 2, 4, 24, 16
+    x = _bf16(torch.randn(B, H, L, dh))
+    y = ops.hip_ops().out_repack(x, False)
+    expect = x.permute(0, 2, 1, 3).reshape(B, L, H * dh)
+    assert torch.equal(y, expect)
+    back = ops.hip_ops().out_repack_bwd(y, H)
+    assert torch.equal(back, x)

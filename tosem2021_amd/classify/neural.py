@@ -69,17 +69,31 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
     dev = torch.device(device) if device else (
         torch.device("cuda") if torch.cuda.is_available() else
         torch.device("cpu"))
-    df = load_taxonomy(taxonomy_path)
-    full = TaxonomyDataset.from_taxonomy(df)
-    train_ds, val_ds = full.split(val_frac=0.1, seed=seed)
+    if taxonomy_path.endswith(".pt"):
+        full = TaxonomyDataset.from_tensor_file(taxonomy_path)
+    else:
+        df = load_taxonomy(taxonomy_path)
+        full = TaxonomyDataset.from_taxonomy(df)
 
     base = CONFIGS[model]
     cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
+    if full._tokens is not None and int(full._tokens.max()) >= cfg.vocab_size:
+        # prepared file was tokenized for a bigger vocab (e.g. mltc-base);
+        # fold ids into this config's hash space deterministically
+        from tosem2021_amd.models.tokenizer import N_RESERVED
+        t = full._tokens
+        big = t >= N_RESERVED
+        full._tokens = torch.where(
+            big, (t - N_RESERVED) % (cfg.vocab_size - N_RESERVED) + N_RESERVED,
+            t)
+    train_ds, val_ds = full.split(val_frac=0.1, seed=seed)
     tok = CodeTokenizer(cfg.vocab_size)
     tcfg = TrainConfig(model=model, lr=lr, warmup_steps=min(50, steps // 10),
                        total_steps=steps, ckpt_dir=ckpt_dir,
                        dtype="bf16" if dev.type == "cuda" else "f32")
     trainer = Trainer(tcfg, device=dev, model_cfg=cfg)
+    trainer.model.set_pos_weights(
+        {k: v.to(dev) for k, v in train_ds.pos_weights().items()})
     if resume and ckpt_dir:
         trainer.load_or_init()
 
@@ -104,6 +118,9 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                   f"methodAcc {ev['method_accuracy']:.3f}")
     train_time = time.time() - t0
     ev = evaluate(trainer, val_ds, tok, seq)
+    ev_lo = evaluate(trainer, val_ds, tok, seq, threshold=0.3)
+    ev["strategy_micro_f1_t0.3"] = ev_lo["strategy_micro_f1"]
+    ev["property_micro_f1_t0.3"] = ev_lo["property_micro_f1"]
     if ckpt_dir:
         trainer.save()
     return {

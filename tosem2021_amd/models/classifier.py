@@ -95,17 +95,13 @@ class Attention(nn.Module):
 
     def forward(self, x, mask_bias: Optional[torch.Tensor]):
         B, L, D = x.shape
-        # one contiguous repack [3, B, H, L, dh] instead of three strided bmm
-        # operands (strided 4D matmul was materializing batched copies every
-        # layer — profiles/r01_kernel_stats_baseline.md, CatArrayBatchedCopy)
-        qkv = (self.qkv(x).view(B, L, 3, self.n_heads, self.head_dim)
-               .permute(2, 0, 3, 1, 4).contiguous())
-        q, k, v = qkv.unbind(0)                           # [B, H, L, dh]
+        # single gather kernel to bmm-ready [3, B, H, L, dh] (csrc/repack.hip)
+        # instead of torch permute-copies + CatArrayBatchedCopy in backward
+        q, k, v = ops.qkv_repack(self.qkv(x), self.n_heads)  # [B, H, L, dh]
         scores = torch.matmul(q, k.transpose(-1, -2))     # [B, H, L, L]
         p = ops.fused_softmax(scores, mask_bias, self.scale)
         out = torch.matmul(p, v)                          # [B, H, L, dh]
-        out = out.transpose(1, 2).contiguous().view(B, L, D)
-        return self.proj(out)
+        return self.proj(ops.out_repack(out))
 
 
 class FFN(nn.Module):
@@ -205,6 +201,12 @@ class MLTC(nn.Module):
             pooled = x.mean(dim=1)
         return {name: head(pooled) for name, head in self.heads.items()}
 
+    def set_pos_weights(self, weights: Dict[str, torch.Tensor]) -> None:
+        """Per-class positive weights for the multi-label heads (label
+        frequencies in the taxonomy are heavily imbalanced)."""
+        for name, w in weights.items():
+            self.register_buffer(f"_pw_{name}", w.float(), persistent=False)
+
     def loss(self, logits: Dict[str, torch.Tensor],
              labels: Dict[str, torch.Tensor]) -> torch.Tensor:
         total = None
@@ -213,7 +215,9 @@ class MLTC(nn.Module):
                 continue
             lg = lg.float()
             if name in MULTILABEL_HEADS:
-                li = F.binary_cross_entropy_with_logits(lg, labels[name].float())
+                pw = getattr(self, f"_pw_{name}", None)
+                li = F.binary_cross_entropy_with_logits(
+                    lg, labels[name].float(), pos_weight=pw)
             else:
                 li = F.cross_entropy(lg, labels[name])
             total = li if total is None else total + li

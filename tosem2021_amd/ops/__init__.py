@@ -157,6 +157,64 @@ def fused_softmax(scores, mask=None, scale: float = 1.0):
     return _SoftmaxFn.apply(scores.contiguous(), mask, scale)
 
 
+class _QkvRepackFn(torch.autograd.Function):
+    """[B, L, 3*H*dh] -> three contiguous [B, H, L, dh] (bmm-ready q/k/v).
+
+    Backward gathers (dq, dk, dv) straight back to the Linear's layout in one
+    kernel — no stack/cat materialization."""
+
+    @staticmethod
+    def forward(ctx, qkv, n_heads):
+        ctx.n_heads = n_heads
+        if qkv.is_cuda:
+            stacked = hip_ops().qkv_repack(qkv, n_heads, False)
+        else:
+            B, L, _ = qkv.shape
+            dh = qkv.shape[-1] // (3 * n_heads)
+            stacked = (qkv.view(B, L, 3, n_heads, dh).permute(2, 0, 3, 1, 4)
+                       .contiguous())
+        q, k, v = stacked.unbind(0)
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        dq = dq.contiguous(); dk = dk.contiguous(); dv = dv.contiguous()
+        if dq.is_cuda:
+            return hip_ops().qkv_repack_bwd3(dq, dk, dv), None
+        B, H, L, dh = dq.shape
+        g = torch.stack([dq, dk, dv], dim=0)
+        return g.permute(1, 3, 0, 2, 4).reshape(B, L, 3 * H * dh), None
+
+
+def qkv_repack(qkv, n_heads: int):
+    return _QkvRepackFn.apply(qkv.contiguous(), n_heads)
+
+
+class _OutRepackFn(torch.autograd.Function):
+    """[B, H, L, dh] -> [B, L, H*dh] (attention output merge)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.n_heads = x.shape[1]
+        if x.is_cuda:
+            return hip_ops().out_repack(x, False)
+        B, H, L, dh = x.shape
+        return x.permute(0, 2, 1, 3).reshape(B, L, H * dh)
+
+    @staticmethod
+    def backward(ctx, g):
+        g = g.contiguous()
+        if g.is_cuda:
+            return hip_ops().out_repack_bwd(g, ctx.n_heads)
+        B, L, D = g.shape
+        H = ctx.n_heads
+        return g.view(B, L, H, D // H).permute(0, 2, 1, 3).contiguous()
+
+
+def out_repack(x):
+    return _OutRepackFn.apply(x.contiguous())
+
+
 def adamw_step(p, grad, m, v, master, *, lr, beta1=0.9, beta2=0.999, eps=1e-8,
                wd=0.01, step, grad_scale=1.0):
     """Fused AdamW over the flat parameter buffer (see train.py)."""
