@@ -43,7 +43,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=24)
     ap.add_argument("--warmup", type=int, default=6)
-    ap.add_argument("--batch", type=int, default=64, help="per-GPU batch")
+    ap.add_argument("--batch", type=int, default=128, help="per-GPU batch")
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--model", type=str, default="mltc-base")
     ap.add_argument("--bucket-mb", type=int, default=64)

@@ -1,0 +1,230 @@
+#include "hip/hip_runtime.h"
+// Flash-attention forward for gfx950 (CDNA4) — bf16 I/O, f32 online softmax,
+// MFMA (v_mfma_f32_32x32x16_bf16) for QK^T and PV, head_dim = 64.
+//
+// Never materializes the [L, L] score matrix: per 32-row Q block (one wave),
+// iterate 32-key KV tiles with the running-max/denominator recurrence and
+// accumulate O in registers.  Saves the per-row logsumexp for the recompute
+// backward (tosem2021_amd/ops: bwd = bmm-recompute + softmax_bwd).
+//
+// Structure (guide §B "fused attention prefill" adapted):
+//  * swapped QK^T — compute mfma(A=K, B=Q^T) so the score column for one
+//    q-row is LANE-LOCAL (16 regs + the partner lane at lane^32): softmax is
+//    15 in-lane max/sum ops + one __shfl_xor(32) exchange, no LDS round trip.
+//  * P -> PV A-fragments via v_cvt_pk_bf16_f32 pairs + permlane32_swap
+//    (guide T12): 8 cvt_pk + 4 swaps replace any cross-lane scatter.
+//  * K tile staged row-major in LDS with the T2 XOR swizzle
+//    (byte ^= (row&7)<<4) so the fragment ds_read_b128 is ~conflict-free;
+//    V tile staged TRANSPOSED ([64 d][32 kv], 80 B row stride) so the PV
+//    B-fragment is a clean ds_read_b128 as well.
+//  * workgroup = 4 waves = 128 q rows of one (b, h); K/V tiles staged once
+//    per workgroup and consumed by all four waves.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define FA_DH 64
+#define FA_KVB 32
+#define FA_QB 32          // q rows per wave
+#define FA_WAVES 4
+#define FA_BLOCK (FA_WAVES * WAVE)
+#define FA_QWG (FA_WAVES * FA_QB)   // 128 q rows per workgroup
+
+// LDS layout (dynamic, 16-B aligned):
+//  K tile:  [32][64] bf16, row stride 128 B, XOR-swizzled      = 4096 B
+//  Vt tile: [64][32] bf16, row stride 80 B (pad 8)             = 5120 B
+//  alpha:   [4 waves][32] f32                                  =  512 B
+#define K_LDS_BYTES (FA_KVB * 128)
+#define VT_ROW_BYTES 80
+#define VT_LDS_BYTES (FA_DH * VT_ROW_BYTES)
+
+__device__ __forceinline__ int kswz(int row, int byte_off) {
+  return byte_off ^ ((row & 7) << 4);
+}
+
+extern "C" __global__ void __launch_bounds__(FA_BLOCK)
+flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
+                 const short* __restrict__ v, const float* __restrict__ mask,
+                 short* __restrict__ o, float* __restrict__ lse,
+                 int B, int H, int L, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* vt_lds = (short*)(smem + K_LDS_BYTES);      // [64][40] (80 B rows)
+  float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
+  const int half = lane >> 5;
+
+  // grid: x = (b*H + h) * n_qblocks + qb
+  const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
+  int bh = blockIdx.x / n_qblocks;
+  int qb = blockIdx.x % n_qblocks;
+  const int b = bh / H;
+  const long bh_off = (long)bh * L * FA_DH;
+  const int q_base_wg = qb * FA_QWG;
+  const int q_base = q_base_wg + wid * FA_QB;        // this wave's 32 q rows
+  const int my_q = q_base + col;                      // this lane's q row
+  const bool q_valid = my_q < L;
+  const float* mrow = mask ? mask + (long)b * L : nullptr;
+
+  // ---- load Q fragments (once): A/B-operand layout, 4 k-chunks of 16 ----
+  // lane holds Q[my_q][16c + 8*half + j], j=0..8
+  short8_t qf[4];
+  {
+    const short* qr = q + bh_off + (long)(q_valid ? my_q : L - 1) * FA_DH;
+#pragma unroll
+    for (int c = 0; c < 4; ++c)
+      qf[c] = *(const short8_t*)(qr + c * 16 + half * 8);
+  }
+
+  f32x16 o_acc[2];                 // O[d-tile t][16 q-rows], d = 32t + col
+#pragma unroll
+  for (int t = 0; t < 2; ++t) o_acc[t] = (f32x16)(0.f);
+  float m_run = -3.0e38f;
+  float l_run = 0.f;
+
+  const int n_kv = L / FA_KVB;     // host asserts L % 32 == 0
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int kv0 = kt * FA_KVB;
+    // ---- stage K (swizzled) and V^T cooperatively ----
+    __syncthreads();
+    {
+      // K: 256 threads x one short8: row = tid/8 (32 rows), col8 = tid%8
+      int row = tid >> 3, c8 = (tid & 7) * 16;  // byte col
+      short8_t kv8 = *(const short8_t*)(k + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
+      *(short8_t*)((char*)k_lds + row * 128 + kswz(row, c8)) = kv8;
+      // V: same global packet, transposed scatter into vt_lds
+      short8_t vv8 = *(const short8_t*)(v + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = (c8 >> 1) + j;
+        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + row * 2) = vv8[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
+    f32x16 s_acc = (f32x16)(0.f);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      // K fragment: lane holds K[kv=col][16c + 8*half + j] from swizzled LDS
+      int row = col;
+      int byte_off = (16 * c + 8 * half) * 2;
+      short8_t kf = *(const short8_t*)((char*)k_lds + row * 128 +
+                                       kswz(row, byte_off));
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc, 0, 0, 0);
+    }
+
+    // ---- online softmax (per lane: one q column, 16 kv rows) ----
+    float sv[16];
+    float tmax = -3.0e38f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+      float x = s_acc[r] * scale;
+      if (mrow) x += mrow[kv0 + kv_local];
+      sv[r] = x;
+      tmax = fmaxf(tmax, x);
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+    float m_new = fmaxf(m_run, tmax);
+    float a = __expf(m_run - m_new);
+    float rsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      sv[r] = __expf(sv[r] - m_new);
+      rsum += sv[r];
+    }
+    rsum += __shfl_xor(rsum, 32, WAVE);
+    l_run = l_run * a + rsum;
+    m_run = m_new;
+
+    // broadcast alpha(q) to the O accumulator rows via per-wave LDS
+    // (both halves write the same value; wave-internal ds ordering suffices)
+    alpha_lds[wid * 32 + col] = a;
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        o_acc[t][r] *= alpha_lds[wid * 32 + qrow];
+      }
+    }
+
+    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
+    // chunk 0: kv 8*half..+8 from regs 0..7; chunk 1: kv 16+8*half from 8..15
+    short8_t pf[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+      uint4_t u;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int r0 = c * 8 + 2 * i;         // own pair (kv low quad)
+        int r1 = c * 8 + 4 + 2 * i;     // pair the partner half needs (+8)
+        unsigned lo, hi;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(lo)
+            : "v"(sv[r0]), "v"(sv[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(hi)
+            : "v"(sv[r1]), "v"(sv[r1 + 1]));
+        auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
+        // consecutive-kv order: {01, 23, 45, 67} (own pairs i=0,1 first)
+        u[i] = sw[0];
+        u[i + 2] = sw[1];
+      }
+      pf[c] = __builtin_bit_cast(short8_t, u);
+    }
+
+    // ---- PV: O[q, d] += P^T @ V — B-fragment from transposed V tile ----
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        // lane holds V[kv = 16c + 8*half + j][d = 32t + col]
+        int d = 32 * t + col;
+        int kvb = 16 * c + 8 * half;
+        short8_t vf = *(const short8_t*)((char*)vt_lds + d * VT_ROW_BYTES +
+                                         kvb * 2);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
+                                                           o_acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
+  //      store O rows + logsumexp ----
+  if (q_valid && lse != nullptr && half == 0)
+    lse[(long)bh * L + my_q] = m_run + __logf(l_run);
+  alpha_lds[wid * 32 + col] = 1.0f / l_run;
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
+      int qrow = q_base + rloc;
+      if (qrow >= L) continue;
+      o[bh_off + (long)qrow * FA_DH + 32 * t + col] =
+          f32_to_bf16(o_acc[t][r] * alpha_lds[wid * 32 + rloc]);
+    }
+  }
+}
+
+extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
+                                       const void* v, const void* mask,
+                                       void* o, void* lse, int B, int H,
+                                       int L, float scale,
+                                       hipStream_t stream) {
+  int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
+  dim3 grid(B * H * n_qblocks);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
+ hipLaunchKernelGGL(( flash_fwd_kernel), dim3(grid), dim3(FA_BLOCK), shm, stream, 
+      (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
+      (short*)o, (float*)lse, B, H, L, scale);
+  return hipGetLastError();
+}

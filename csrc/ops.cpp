@@ -35,6 +35,10 @@ hipError_t out_repack_launch(const void*, void*, int, int, int, int, int,
                              hipStream_t);
 hipError_t qkv_repack_bwd3_launch(const void*, const void*, const void*,
                                   void*, int, int, int, int, hipStream_t);
+hipError_t flash_fwd_launch(const void*, const void*, const void*, const void*,
+                            void*, void*, int, int, int, float, hipStream_t);
+hipError_t p_from_lse_launch(const void*, const void*, const void*, void*,
+                             long, int, int, float, int, hipStream_t);
 }
 
 namespace {
@@ -270,7 +274,54 @@ torch::Tensor out_repack_bwd(torch::Tensor dgrad, long H) {
   return out;
 }
 
+std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k,
+                                     torch::Tensor v,
+                                     c10::optional<torch::Tensor> mask,
+                                     double scale) {
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  TORCH_CHECK(q.dim() == 4, "q must be [B, H, L, dh]");
+  const long B = q.size(0), H = q.size(1), L = q.size(2), dh = q.size(3);
+  TORCH_CHECK(dh == 64, "flash_fwd is specialized for head_dim 64");
+  TORCH_CHECK(L % 32 == 0, "flash_fwd needs L % 32 == 0");
+  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    TORCH_CHECK(mask->size(0) == B && mask->size(-1) == L, "mask is [B, L]");
+    mptr = mask->data_ptr();
+  }
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, L}, q.options().dtype(torch::kFloat32));
+  CHECK_HIP(flash_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), mptr,
+                             o.data_ptr(), lse.data_ptr(), (int)B, (int)H,
+                             (int)L, (float)scale, cur_stream()));
+  return {o, lse};
+}
+
+torch::Tensor p_from_lse(torch::Tensor scores, c10::optional<torch::Tensor> mask,
+                         torch::Tensor lse, double scale) {
+  check_bf16(scores, "scores"); check_f32(lse, "lse");
+  const int Lk = (int)scores.size(-1);
+  const long n_rows = scores.numel() / Lk;
+  TORCH_CHECK(lse.numel() == n_rows, "lse size mismatch");
+  int H_Lq = 1;
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    H_Lq = (int)(scores.size(1) * scores.size(2));
+    mptr = mask->data_ptr();
+  }
+  auto p = torch::empty_like(scores);
+  int grid = (int)std::min<long>((n_rows + 3) / 4, 2048);
+  CHECK_HIP(p_from_lse_launch(scores.data_ptr(), mptr, lse.data_ptr(),
+                              p.data_ptr(), n_rows, Lk, H_Lq, (float)scale,
+                              grid, cur_stream()));
+  return p;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
+  m.def("p_from_lse", &p_from_lse, "probabilities from saved logsumexp");
   m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");
   m.def("qkv_repack_bwd3", &qkv_repack_bwd3, "qkv repack bwd from dq,dk,dv");
   m.def("out_repack", &out_repack, "attention output merge");

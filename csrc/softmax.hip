@@ -193,3 +193,46 @@ hipError_t softmax_bwd_launch(const void* dp, const void* p, void* ds,
 }
 
 }  // extern "C"
+
+// P = exp(scale * S + mask_bias - lse_row): elementwise probability
+// recompute from the flash forward's logsumexp (flash backward path).
+extern "C" __global__ void __launch_bounds__(SM_BLOCK)
+p_from_lse_kernel(const short* __restrict__ s_in, const float* __restrict__ mask,
+                  const float* __restrict__ lse, short* __restrict__ p_out,
+                  long n_rows, int Lk, int H_Lq, float scale) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  for (long row = blockIdx.x * SM_WPB + wid; row < n_rows;
+       row += (long)gridDim.x * SM_WPB) {
+    const short* sr = s_in + row * Lk;
+    short* pr = p_out + row * Lk;
+    const float* mrow = mask ? mask + (row / H_Lq) * Lk : nullptr;
+    const float l = lse[row];
+    for (int i = lane * 8; i < Lk; i += WAVE * 8) {
+      short8_t x = *(const short8_t*)(sr + i);
+      float4_t m0, m1;
+      if (mrow) {
+        m0 = *(const float4_t*)(mrow + i);
+        m1 = *(const float4_t*)(mrow + i + 4);
+      }
+      short8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32(x[j]) * scale - l;
+        if (mrow) f += j < 4 ? m0[j] : m1[j - 4];
+        o[j] = f32_to_bf16(__expf(f));
+      }
+      *(short8_t*)(pr + i) = o;
+    }
+  }
+}
+
+extern "C" hipError_t p_from_lse_launch(const void* s, const void* mask,
+                                        const void* lse, void* p, long n_rows,
+                                        int Lk, int H_Lq, float scale,
+                                        int grid, hipStream_t stream) {
+  p_from_lse_kernel<<<grid, SM_BLOCK, 0, stream>>>(
+      (const short*)s, (const float*)mask, (const float*)lse, (short*)p,
+      n_rows, Lk, H_Lq, scale);
+  return hipGetLastError();
+}

@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "csrc/softmax.hip",
         "csrc/adamw.hip",
         "csrc/repack.hip",
+        "csrc/flash_attn.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

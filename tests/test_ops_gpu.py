@@ -178,3 +178,75 @@ def test_out_repack_roundtrip():
     assert torch.equal(y, expect)
     back = ops.hip_ops().out_repack_bwd(y, H)
     assert torch.equal(back, x)
+
+
+@pytest.mark.parametrize("B,H,L", [(2, 3, 128), (1, 2, 512), (2, 2, 96)])
+def test_flash_fwd_vs_reference(B, H, L):
+    torch.manual_seed(10)
+    q = _bf16(torch.randn(B, H, L, 64))
+    k = _bf16(torch.randn(B, H, L, 64))
+    v = _bf16(torch.randn(B, H, L, 64))
+    mask = torch.zeros(B, L)
+    mask[:, L - L // 4:] = -1e9
+    mg = mask.cuda().contiguous()
+    scale = 1.0 / 8.0
+    o, lse = ops.hip_ops().flash_fwd(q, k, v, mg, scale)
+    oe, le = ref.flash_attention_fwd(q.float().cpu(), k.float().cpu(),
+                                     v.float().cpu(), mask, scale)
+    assert torch.allclose(o.float().cpu(), oe.float(), atol=3e-2, rtol=3e-2), \
+        (o.float().cpu() - oe.float()).abs().max()
+    assert torch.allclose(lse.cpu(), le, atol=2e-2, rtol=1e-3)
+
+
+def test_flash_fwd_no_mask():
+    torch.manual_seed(11)
+    q = _bf16(torch.randn(2, 2, 64, 64) * 2)
+    k = _bf16(torch.randn(2, 2, 64, 64) * 2)
+    v = _bf16(torch.randn(2, 2, 64, 64))
+    o, lse = ops.hip_ops().flash_fwd(q, k, v, None, 0.125)
+    oe, le = ref.flash_attention_fwd(q.float().cpu(), k.float().cpu(),
+                                     v.float().cpu(), None, 0.125)
+    assert torch.allclose(o.float().cpu(), oe.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(lse.cpu(), le, atol=2e-2, rtol=1e-3)
+
+
+def test_flash_backward_matches_bmm_path():
+    """End-to-end grads: flash fwd + recompute bwd == bmm+softmax autograd."""
+    torch.manual_seed(12)
+    B, H, L, dh = 2, 2, 96, 64
+    scale = 1.0 / 8.0
+    base = {n: _bf16(torch.randn(B, H, L, dh)) for n in "qkv"}
+    mask = torch.zeros(B, L)
+    mask[:, 80:] = -1e9
+    mg = mask.cuda().contiguous()
+    do = _bf16(torch.randn(B, H, L, dh))
+
+    fa = {n: t.clone().requires_grad_(True) for n, t in base.items()}
+    o1 = ops.flash_attention(fa["q"], fa["k"], fa["v"], mg, scale)
+    o1.backward(do)
+
+    bm = {n: t.clone().requires_grad_(True) for n, t in base.items()}
+    scores = torch.matmul(bm["q"], bm["k"].transpose(-1, -2))
+    p = ops.fused_softmax(scores, mg, scale)
+    o2 = torch.matmul(p, bm["v"])
+    o2.backward(do)
+
+    assert torch.allclose(o1.float(), o2.float(), atol=4e-2, rtol=4e-2)
+    for n in "qkv":
+        a, b = fa[n].grad.float(), bm[n].grad.float()
+        assert torch.allclose(a, b, atol=6e-2, rtol=6e-2), \
+            f"d{n}: {(a-b).abs().max()}"
+
+
+def test_p_from_lse_rows_normalized():
+    torch.manual_seed(13)
+    B, H, L = 2, 2, 128
+    q = _bf16(torch.randn(B, H, L, 64))
+    k = _bf16(torch.randn(B, H, L, 64))
+    v = _bf16(torch.randn(B, H, L, 64))
+    scale = 0.125
+    _, lse = ops.hip_ops().flash_fwd(q, k, v, None, scale)
+    s = torch.matmul(q, k.transpose(-1, -2)).contiguous()
+    p = ops.hip_ops().p_from_lse(s, None, lse, scale)
+    sums = p.float().sum(-1)
+    assert torch.allclose(sums, torch.ones_like(sums), atol=3e-2)

@@ -215,6 +215,52 @@ def out_repack(x):
     return _OutRepackFn.apply(x.contiguous())
 
 
+class _FlashAttentionFn(torch.autograd.Function):
+    """Flash attention: MFMA forward (csrc/flash_attn.hip, O(L) memory, saves
+    logsumexp), recompute backward (bmm + p_from_lse + softmax_bwd)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, mask, scale):
+        if q.is_cuda:
+            o, lse = hip_ops().flash_fwd(q, k, v, mask, scale)
+        else:
+            o, lse = reference.flash_attention_fwd(q, k, v, mask, scale)
+        ctx.save_for_backward(q, k, v, lse)
+        ctx.mask = mask
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, lse = ctx.saved_tensors
+        mask, scale = ctx.mask, ctx.scale
+        do = do.contiguous()
+        s = torch.matmul(q, k.transpose(-1, -2))
+        if q.is_cuda:
+            p = hip_ops().p_from_lse(s.contiguous(), mask, lse, scale)
+        else:
+            p = reference.p_from_lse(s, mask, lse, scale)
+        dp = torch.matmul(do, v.transpose(-1, -2))
+        if q.is_cuda:
+            dsc = hip_ops().softmax_bwd(dp.contiguous(), p, scale)
+        else:
+            dsc = reference.softmax_bwd(dp, p, scale)
+        dq = torch.matmul(dsc, k)
+        dk = torch.matmul(dsc.transpose(-1, -2), q)
+        dv = torch.matmul(p.transpose(-1, -2), do)
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q, k, v, mask=None, scale: float = 1.0):
+    """q/k/v: [B, H, L, 64] bf16 contiguous, L % 32 == 0; mask: [B, L] f32."""
+    return _FlashAttentionFn.apply(q.contiguous(), k.contiguous(),
+                                   v.contiguous(), mask, scale)
+
+
+def flash_supported(head_dim: int, L: int) -> bool:
+    return head_dim == 64 and L % 32 == 0
+
+
 def adamw_step(p, grad, m, v, master, *, lr, beta1=0.9, beta2=0.999, eps=1e-8,
                wd=0.01, step, grad_scale=1.0):
     """Fused AdamW over the flat parameter buffer (see train.py)."""

@@ -87,6 +87,24 @@ def softmax_bwd(dp: torch.Tensor, p: torch.Tensor, scale: float) -> torch.Tensor
     return (scale * pf * (dpf - dot)).to(p.dtype)
 
 
+def flash_attention_fwd(q, k, v, mask, scale):
+    """fp32 reference of the flash forward: returns (o in q.dtype, lse f32)."""
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if mask is not None:
+        s = s + mask.float().view(mask.shape[0], 1, 1, -1)
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, v.float())
+    return o.to(q.dtype), lse
+
+
+def p_from_lse(scores, mask, lse, scale):
+    s = scores.float() * scale
+    if mask is not None:
+        s = s + mask.float().view(mask.shape[0], 1, 1, -1)
+    return torch.exp(s - lse.unsqueeze(-1)).to(scores.dtype)
+
+
 def adamw_step(
     p: torch.Tensor,
     grad: torch.Tensor,
