@@ -133,3 +133,78 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
         **{k: (round(v, 4) if isinstance(v, float) else v)
            for k, v in ev.items()},
     }
+
+
+@torch.no_grad()
+def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
+                     model: str = "mltc-base", seq: int = 256,
+                     batch: int = 64, threshold: float = 0.5,
+                     device: Optional[str] = None) -> str:
+    """Label the rows of a (mined) taxonomy CSV with a trained MLTC model and
+    rewrite the 41-column CSV with the predicted labels."""
+    from tosem2021_amd.extract.schema import (
+        METHODS, PROPERTIES, STAGES, STRATEGIES, STRATEGY_TO_COLUMNS,
+        TestCaseRow)
+    from tosem2021_amd.pipeline import write_taxonomy_csv
+
+    dev = torch.device(device) if device else (
+        torch.device("cuda") if torch.cuda.is_available() else
+        torch.device("cpu"))
+    df = load_taxonomy(taxonomy_path)
+    base = CONFIGS[model]
+    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
+    tok = CodeTokenizer(cfg.vocab_size)
+    tcfg = TrainConfig(model=model, ckpt_dir=ckpt_dir,
+                       dtype="bf16" if dev.type == "cuda" else "f32")
+    trainer = Trainer(tcfg, device=dev, model_cfg=cfg)
+    assert trainer.load_or_init(), f"no checkpoint found in {ckpt_dir}"
+    trainer.model.eval()
+
+    texts = (df["Labels"].astype(str) + " | " +
+             df["Component"].astype(str)).tolist()
+    rows = []
+    for lo in range(0, len(texts), batch):
+        chunk = texts[lo:lo + batch]
+        toks, mask = tok.encode_batch(chunk, seq, device=dev)
+        logits = trainer.model(toks, mask)
+        sp = torch.sigmoid(logits["strategy"].float()) > threshold
+        pp = torch.sigmoid(logits["property"].float()) > threshold
+        stg = logits["stage"].float().argmax(-1)
+        mth = logits["method"].float().argmax(-1)
+        for i in range(len(chunk)):
+            src = df.iloc[lo + i]
+            row = TestCaseRow(
+                index=lo + i + 1, labels=str(src["Labels"]),
+                component=str(src["Component"]), repo=str(src["Repo"]),
+                category=STAGES[int(stg[i])], category2=STAGES[int(stg[i])])
+            flags = {}
+            for j, name in enumerate(STRATEGIES):
+                if not bool(sp[i, j]):
+                    continue
+                spec = STRATEGY_TO_COLUMNS[name]
+                flags.update(spec.get("flags", {}))
+                if "error_type" in spec:
+                    row.error_type = spec["error_type"]
+                if "approximation_type" in spec:
+                    row.approximation_type = spec["approximation_type"]
+                if "checks_type" in spec:
+                    row.checks_type = spec["checks_type"]
+            m = METHODS[int(mth[i])]
+            if m == "regression":
+                flags["regression"] = 1
+            elif m == "integration":
+                flags["Integration"] = 1
+            elif m == "end_to_end":
+                flags["end_to_end"] = 1
+            if not flags:
+                flags["None_above"] = 1
+            row.flags = flags
+            props = [PROPERTIES[j] for j in range(len(PROPERTIES))
+                     if bool(pp[i, j])]
+            if props:
+                row.model = props[0]
+            if len(props) > 1:
+                row.data = props[1]
+            rows.append(row)
+    trainer.model.train()
+    return write_taxonomy_csv(rows, out_csv)

@@ -66,6 +66,14 @@ def cmd_report(args):
     print(summary_report(df))
 
 
+def cmd_classify(args):
+    from tosem2021_amd.classify.neural import apply_classifier
+    out = apply_classifier(args.ckpt_dir, args.taxonomy, args.out,
+                           model=args.model, seq=args.seq,
+                           threshold=args.threshold)
+    print(f"wrote {out}")
+
+
 def cmd_train(args):
     from tosem2021_amd.classify.neural import train_classifier
     res = train_classifier(
@@ -107,6 +115,15 @@ def main(argv=None):
     p = sub.add_parser("report", help="print a taxonomy summary report")
     p.add_argument("--taxonomy", required=True)
     p.set_defaults(fn=cmd_report)
+
+    p = sub.add_parser("classify", help="label a taxonomy with a trained model")
+    p.add_argument("--taxonomy", required=True)
+    p.add_argument("--ckpt-dir", required=True)
+    p.add_argument("--out", required=True)
+    p.add_argument("--model", default="mltc-base")
+    p.add_argument("--seq", type=int, default=256)
+    p.add_argument("--threshold", type=float, default=0.5)
+    p.set_defaults(fn=cmd_classify)
 
     p = sub.add_parser("train", help="train the MLTC classifier on a taxonomy")
     p.add_argument("--taxonomy", required=True)

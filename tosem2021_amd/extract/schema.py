@@ -159,6 +159,37 @@ APPROX_TYPE_TO_STRATEGY: Dict[str, str] = {
 }
 
 
+# Inverse mapping: strategy label -> the taxonomy columns that encode it
+# (used when writing model-predicted labels back into the 41-column schema).
+STRATEGY_TO_COLUMNS = {
+    "status_analysis": {"flags": {"status_test": 1}},
+    "negative_test": {"flags": {"negative_test": 1}},
+    "logical_condition": {"flags": {"logical_expression": 1}},
+    "Null_pointer": {"flags": {"null_pointer": 1}},
+    "value_range_analysis": {"flags": {"value_range": 1}},
+    "value_error": {"flags": {"error_handling": 1}, "error_type": "ValueError"},
+    "runtime_error": {"flags": {"error_handling": 1}, "error_type": "RuntimeError"},
+    "memory_error": {"flags": {"error_handling": 1}, "error_type": "MemoryError"},
+    "type_error": {"flags": {"error_handling": 1}, "error_type": "TypeError"},
+    "import_error": {"flags": {"error_handling": 1}, "error_type": "ImportError"},
+    "key_error": {"flags": {"error_handling": 1}, "error_type": "KeyError"},
+    "AssertionError": {"flags": {"error_handling": 1},
+                       "error_type": "AssertionError"},
+    "FileError": {"flags": {"error_handling": 1}, "error_type": "FileError"},
+    "NotImplementedError": {"flags": {"error_handling": 1},
+                            "error_type": "NotImplementedError"},
+    "absolute_relative_tolerence": {
+        "flags": {"Approximation": 1},
+        "approximation_type": "absolute_relative_tolerence"},
+    "rounding_tolence": {"flags": {"Approximation": 1},
+                         "approximation_type": "rounding_tolence"},
+    "error_bounding": {"flags": {"Approximation": 1},
+                       "approximation_type": "error_bounding"},
+    "instance_check": {"checks_type": "instance_check"},
+    "sub_set_checks": {"checks_type": "sub_set_checks"},
+}
+
+
 @dataclass
 class TestCaseRow:
     """One extracted/labeled test case or assertion (one taxonomy row)."""
