@@ -68,3 +68,15 @@ __device__ __forceinline__ float gelu_tanh_grad(float x) {
 
 // Kernel files expose C-linkage launchers returning hipError_t; the torch
 // binding layer (ops.cpp) is the only place that includes torch headers.
+
+// Place `group` consecutive block ids on the same XCD (dispatcher maps block
+// b -> XCD b % 8; guide T1).  Bijective when (n_blocks / group) % 8 == 0;
+// falls back to identity otherwise.  Performance-only (G16: never rely on
+// placement for correctness).
+__device__ __forceinline__ int xcd_group_remap(int bid, int n_blocks,
+                                               int group) {
+  int per = n_blocks / group;
+  if (n_blocks % group != 0 || per % 8 != 0) return bid;
+  int g = bid / group, r = bid % group;
+  return g + r * per;
+}

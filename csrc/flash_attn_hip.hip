@@ -59,10 +59,12 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
   const int half = lane >> 5;
 
-  // grid: x = (b*H + h) * n_qblocks + qb
+  // grid: x = (b*H + h) * n_qblocks + qb; remapped so one (b,h)'s q-blocks
+  // share an XCD and re-read K/V from that XCD's L2 (guide T1)
   const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
-  int bh = blockIdx.x / n_qblocks;
-  int qb = blockIdx.x % n_qblocks;
+  int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
+  int bh = bid / n_qblocks;
+  int qb = bid % n_qblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
   const int q_base_wg = qb * FA_QWG;
@@ -226,5 +228,180 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
  hipLaunchKernelGGL(( flash_fwd_kernel), dim3(grid), dim3(FA_BLOCK), shm, stream, 
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale);
+  return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
+// Flash backward, stage 1: fused recompute of P^T and dS^T.
+//
+//   P^T[kv, q]  = exp(scale * S^T + mask_bias[kv] - lse[q])
+//   dP^T[kv, q] = V @ dO^T
+//   dS^T[kv, q] = scale * P^T * (dP^T - D[q]),  D = rowsum(dO * O)
+//
+// One MFMA pass recomputes both matmul tiles (S^T = K @ Q^T, dP^T = V @ dO^T)
+// from registers + LDS-staged Q/dO tiles; P^T and dS^T are written in the
+// [B, H, L_kv, L_q] layout (coalesced row stores), and the remaining grads
+// are three library bmms in the autograd wrapper:
+//   dV = P^T @ dO,  dK = dS^T @ Q,  dQ = (dS^T)^T @ K.
+// This replaces the unfused recompute chain (QK bmm + p_from_lse +
+// softmax_bwd) with one kernel.
+//
+// Each wave owns one 32-row KV block (K/V fragments live in registers for the
+// whole kernel); the workgroup's 4 waves share the staged Q/dO tiles.
+
+extern "C" __global__ void __launch_bounds__(FA_BLOCK)
+flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
+                    const short* __restrict__ v, const short* __restrict__ dout,
+                    const float* __restrict__ mask,
+                    const float* __restrict__ lse,
+                    const float* __restrict__ ddot,
+                    short* __restrict__ p_t, short* __restrict__ ds_t,
+                    int B, int H, int L, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* q_lds = (short*)smem;                        // swizzled [32][64]
+  short* do_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int col = lane & 31;        // q column of this lane
+  const int half = lane >> 5;
+
+  const int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;   // 128 kv rows per WG
+  int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_kvblocks);
+  int bh = bid / n_kvblocks;
+  int kb = bid % n_kvblocks;
+  const int b = bh / H;
+  const long bh_off = (long)bh * L * FA_DH;
+  const long bh_sq = (long)bh * L * L;
+  const int kv_base = kb * FA_QWG + wid * FA_KVB;     // this wave's 32 kv rows
+  const int my_kv = kv_base + col;
+  const bool kv_valid = my_kv < L;
+  const float* mrow = mask ? mask + (long)b * L : nullptr;
+
+  // K and V fragments for this wave's kv block (resident all kernel)
+  short8_t kf[4], vf[4];
+  {
+    const short* kr = k + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
+    const short* vr = v + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      kf[c] = *(const short8_t*)(kr + c * 16 + half * 8);
+      vf[c] = *(const short8_t*)(vr + c * 16 + half * 8);
+    }
+  }
+  // per-reg mask bias of this wave's kv rows (kv = kv_base + crow(r, half))
+  float mbias[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
+    mbias[r] = mrow ? mrow[min(kv_base + kvl, L - 1)] : 0.f;
+  }
+
+  const int n_q = L / FA_KVB;
+  for (int qt = 0; qt < n_q; ++qt) {
+    const int q0 = qt * 32;
+    __syncthreads();
+    {
+      // stage Q and dO q-tiles, row-major XOR-swizzled (as fwd stages K)
+      int row = tid >> 3, c8 = (tid & 7) * 16;
+      short8_t qv8 = *(const short8_t*)(q + bh_off + (long)(q0 + row) * FA_DH +
+                                        (c8 >> 1));
+      *(short8_t*)((char*)q_lds + row * 128 + kswz(row, c8)) = qv8;
+      short8_t dv8 = *(const short8_t*)(dout + bh_off +
+                                        (long)(q0 + row) * FA_DH + (c8 >> 1));
+      *(short8_t*)((char*)do_lds + row * 128 + kswz(row, c8)) = dv8;
+    }
+    __syncthreads();
+
+    // S^T = K @ Q^T and dP^T = V @ dO^T (B-fragments from the staged tiles:
+    // lane reads row q=col, feature chunk 16c + 8*half)
+    f32x16 s_acc = (f32x16)(0.f);
+    f32x16 dp_acc = (f32x16)(0.f);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int byte_off = (16 * c + 8 * half) * 2;
+      short8_t qfrag = *(const short8_t*)((char*)q_lds + col * 128 +
+                                          kswz(col, byte_off));
+      short8_t dofrag = *(const short8_t*)((char*)do_lds + col * 128 +
+                                           kswz(col, byte_off));
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[c], qfrag, s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[c], dofrag, dp_acc,
+                                                       0, 0, 0);
+    }
+
+    // elementwise: P^T = exp(scale*S^T + bias - lse[q]); dS^T = scale*P*(dP-D)
+    const int my_q = q0 + col;
+    const float l_q = lse[(long)bh * L + my_q];
+    const float d_q = ddot[(long)bh * L + my_q];
+    short8_t p_pack[2], ds_pack[2];
+#pragma unroll
+    for (int rr = 0; rr < 16; rr += 2) {
+      float p0 = __expf(s_acc[rr] * scale + mbias[rr] - l_q);
+      float p1 = __expf(s_acc[rr + 1] * scale + mbias[rr + 1] - l_q);
+      float g0 = scale * p0 * (dp_acc[rr] - d_q);
+      float g1 = scale * p1 * (dp_acc[rr + 1] - d_q);
+      int slot = rr >> 1;  // 8 packed pairs
+      p_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(p0);
+      p_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(p1);
+      ds_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(g0);
+      ds_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(g1);
+    }
+    // store: rows kv = kv_base + crow(2*slot..), cols q0 + col.
+    // Row pairs (2*slot, 2*slot+1) are CONSECUTIVE kv rows, so store the two
+    // bf16 of a pair as separate row stores (2 B per lane, coalesced per row).
+#pragma unroll
+    for (int rr = 0; rr < 16; ++rr) {
+      int kvl = (rr & 3) + 8 * (rr >> 2) + 4 * half;
+      long off = bh_sq + (long)(kv_base + kvl) * L + q0 + col;
+      int slot = rr >> 1;
+      p_t[off] = p_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+      ds_t[off] = ds_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+    }
+  }
+}
+
+extern "C" hipError_t flash_bwd_ds_launch(
+    const void* q, const void* k, const void* v, const void* dout,
+    const void* mask, const void* lse, const void* ddot, void* p_t,
+    void* ds_t, int B, int H, int L, float scale, hipStream_t stream) {
+  int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;
+  dim3 grid(B * H * n_kvblocks);
+  size_t shm = 2 * K_LDS_BYTES;
+ hipLaunchKernelGGL(( flash_bwd_ds_kernel), dim3(grid), dim3(FA_BLOCK), shm, stream, 
+      (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
+      (const float*)mask, (const float*)lse, (const float*)ddot, (short*)p_t,
+      (short*)ds_t, B, H, L, scale);
+  return hipGetLastError();
+}
+
+// D = rowsum(dO * O) per (b, h, q) row — one wave per 4 rows (dh = 64).
+extern "C" __global__ void __launch_bounds__(256)
+fa_dot_kernel(const short* __restrict__ dout, const short* __restrict__ o,
+              float* __restrict__ ddot, long n_rows) {
+  // lane j of quarter-wave handles 4 bf16 (dh=64 -> 16 lanes x 4 elems)
+  long row = ((long)blockIdx.x * 256 + threadIdx.x) >> 4;
+  if (row >= n_rows) return;
+  int sub = threadIdx.x & 15;
+  const short* dr = dout + row * FA_DH + sub * 4;
+  const short* orow = o + row * FA_DH + sub * 4;
+  short4_t a = *(const short4_t*)dr;
+  short4_t c = *(const short4_t*)orow;
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) s += bf16_to_f32(a[j]) * bf16_to_f32(c[j]);
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 16);
+  if (sub == 0) ddot[row] = s;
+}
+
+extern "C" hipError_t fa_dot_launch(const void* dout, const void* o,
+                                    void* ddot, long n_rows,
+                                    hipStream_t stream) {
+  long total_threads = n_rows * 16;
+  int grid = (int)((total_threads + 255) / 256);
+ hipLaunchKernelGGL(( fa_dot_kernel), dim3(grid), dim3(256), 0, stream, (const short*)dout, (const short*)o,
+                                          (float*)ddot, n_rows);
   return hipGetLastError();
 }

@@ -39,6 +39,11 @@ hipError_t flash_fwd_launch(const void*, const void*, const void*, const void*,
                             void*, void*, int, int, int, float, hipStream_t);
 hipError_t p_from_lse_launch(const void*, const void*, const void*, void*,
                              long, int, int, float, int, hipStream_t);
+hipError_t flash_bwd_ds_launch(const void*, const void*, const void*,
+                               const void*, const void*, const void*,
+                               const void*, void*, void*, int, int, int,
+                               float, hipStream_t);
+hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
 }
 
 namespace {
@@ -319,8 +324,44 @@ torch::Tensor p_from_lse(torch::Tensor scores, c10::optional<torch::Tensor> mask
   return p;
 }
 
+torch::Tensor fa_dot(torch::Tensor dout, torch::Tensor o) {
+  check_bf16(dout, "dout"); check_bf16(o, "o");
+  const long n_rows = dout.numel() / dout.size(-1);
+  TORCH_CHECK(dout.size(-1) == 64, "fa_dot expects head_dim 64");
+  auto d = torch::empty({n_rows}, dout.options().dtype(torch::kFloat32));
+  CHECK_HIP(fa_dot_launch(dout.data_ptr(), o.data_ptr(), d.data_ptr(), n_rows,
+                          cur_stream()));
+  return d.view({dout.size(0), dout.size(1), dout.size(2)});
+}
+
+std::vector<torch::Tensor> flash_bwd_ds(torch::Tensor q, torch::Tensor k,
+                                        torch::Tensor v, torch::Tensor dout,
+                                        c10::optional<torch::Tensor> mask,
+                                        torch::Tensor lse, torch::Tensor ddot,
+                                        double scale) {
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(dout, "dout"); check_f32(lse, "lse"); check_f32(ddot, "ddot");
+  const long B = q.size(0), H = q.size(1), L = q.size(2);
+  TORCH_CHECK(q.size(3) == 64 && L % 32 == 0, "dh=64 and L%32==0 required");
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    mptr = mask->data_ptr();
+  }
+  auto p_t = torch::empty({B, H, L, L}, q.options());
+  auto ds_t = torch::empty({B, H, L, L}, q.options());
+  CHECK_HIP(flash_bwd_ds_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                                dout.data_ptr(), mptr, lse.data_ptr(),
+                                ddot.data_ptr(), p_t.data_ptr(),
+                                ds_t.data_ptr(), (int)B, (int)H, (int)L,
+                                (float)scale, cur_stream()));
+  return {p_t, ds_t};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
+  m.def("flash_bwd_ds", &flash_bwd_ds, "fused flash bwd: P^T and dS^T tiles");
+  m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
   m.def("p_from_lse", &p_from_lse, "probabilities from saved logsumexp");
   m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");
   m.def("qkv_repack_bwd3", &qkv_repack_bwd3, "qkv repack bwd from dq,dk,dv");

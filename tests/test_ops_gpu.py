@@ -250,3 +250,32 @@ def test_p_from_lse_rows_normalized():
     p = ops.hip_ops().p_from_lse(s, None, lse, scale)
     sums = p.float().sum(-1)
     assert torch.allclose(sums, torch.ones_like(sums), atol=3e-2)
+
+
+def test_flash_bwd_ds_vs_reference():
+    torch.manual_seed(14)
+    B, H, L = 2, 2, 128
+    scale = 0.125
+    q = _bf16(torch.randn(B, H, L, 64))
+    k = _bf16(torch.randn(B, H, L, 64))
+    v = _bf16(torch.randn(B, H, L, 64))
+    do = _bf16(torch.randn(B, H, L, 64))
+    mask = torch.zeros(B, L)
+    mask[:, 100:] = -1e9
+    mg = mask.cuda().contiguous()
+    o, lse = ops.hip_ops().flash_fwd(q, k, v, mg, scale)
+    ddot = ops.hip_ops().fa_dot(do, o)
+    dd_ref = (do.float() * o.float()).sum(-1)
+    assert torch.allclose(ddot, dd_ref, atol=2e-2, rtol=2e-2)
+    p_t, ds_t = ops.hip_ops().flash_bwd_ds(q, k, v, do, mg, lse, ddot, scale)
+    # reference
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale \
+        + mask.cuda().view(B, 1, 1, L)
+    p_ref = torch.exp(s - lse.unsqueeze(-1))
+    dp_ref = torch.matmul(do.float(), v.float().transpose(-1, -2))
+    ds_ref = scale * p_ref * (dp_ref - dd_ref.unsqueeze(-1))
+    assert torch.allclose(p_t.float().transpose(-1, -2), p_ref, atol=2e-2,
+                          rtol=3e-2)
+    assert torch.allclose(ds_t.float().transpose(-1, -2), ds_ref, atol=2e-2,
+                          rtol=3e-2), \
+        (ds_t.float().transpose(-1, -2) - ds_ref).abs().max()

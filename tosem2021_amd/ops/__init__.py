@@ -225,26 +225,29 @@ class _FlashAttentionFn(torch.autograd.Function):
             o, lse = hip_ops().flash_fwd(q, k, v, mask, scale)
         else:
             o, lse = reference.flash_attention_fwd(q, k, v, mask, scale)
-        ctx.save_for_backward(q, k, v, lse)
+        ctx.save_for_backward(q, k, v, o, lse)
         ctx.mask = mask
         ctx.scale = scale
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, lse = ctx.saved_tensors
+        q, k, v, o, lse = ctx.saved_tensors
         mask, scale = ctx.mask, ctx.scale
         do = do.contiguous()
+        if q.is_cuda:
+            # fused recompute: one MFMA kernel emits P^T and dS^T directly
+            ddot = hip_ops().fa_dot(do, o)
+            p_t, ds_t = hip_ops().flash_bwd_ds(q, k, v, do, mask, lse, ddot,
+                                               scale)
+            dv = torch.matmul(p_t, do)  # [kv,q] @ [q,d]
+            dk = torch.matmul(ds_t, q)
+            dq = torch.matmul(ds_t.transpose(-1, -2), k)
+            return dq, dk, dv, None, None
         s = torch.matmul(q, k.transpose(-1, -2))
-        if q.is_cuda:
-            p = hip_ops().p_from_lse(s.contiguous(), mask, lse, scale)
-        else:
-            p = reference.p_from_lse(s, mask, lse, scale)
+        p = reference.p_from_lse(s, mask, lse, scale)
         dp = torch.matmul(do, v.transpose(-1, -2))
-        if q.is_cuda:
-            dsc = hip_ops().softmax_bwd(dp.contiguous(), p, scale)
-        else:
-            dsc = reference.softmax_bwd(dp, p, scale)
+        dsc = reference.softmax_bwd(dp, p, scale)
         dq = torch.matmul(dsc, k)
         dk = torch.matmul(dsc.transpose(-1, -2), q)
         dv = torch.matmul(p.transpose(-1, -2), do)
