@@ -41,7 +41,7 @@ def test_rule_agreement_floor(ref_taxonomy):
     from tosem2021_amd.classify.agreement import evaluate_rules_on_taxonomy
     res = evaluate_rules_on_taxonomy(ref_taxonomy)
     # documented calibration floor (classify/rules.py); regressions fail here
-    assert res["strategy_micro_f1"] > 0.35, res["strategy_micro_f1"]
+    assert res["strategy_micro_f1"] > 0.55, res["strategy_micro_f1"]
     assert res["method_accuracy"] > 0.95, res["method_accuracy"]
 
 

@@ -39,7 +39,14 @@ def test_raises_value_error():
     assert row.flags.get("error_handling") == 1
     assert row.error_type == "ValueError"
     assert "value_error" in row.strategies()
-    assert "negative_test" in row.strategies()
+
+
+def test_negative_test_is_negated_assertion():
+    # calibrated semantics: the study's negative_test = asserting the negated
+    # condition (EXPECT_FALSE / assertFalse / assertNot*)
+    assert "negative_test" in classify_text("assertFalse(model.ready())").strategies()
+    assert "negative_test" in classify_text("EXPECT_FALSE(queue.empty())").strategies()
+    assert "negative_test" not in classify_text("assertEqual(a, b)").strategies()
 
 
 def test_instance_and_subset_checks():

@@ -26,14 +26,16 @@ from tosem2021_amd.extract.schema import (STAGES, STRATEGIES, TestCaseRow)
 ERROR_NAME_MAP = [
     (re.compile(r"ValueError", re.I), "ValueError", "value_error"),
     (re.compile(r"Runtime_?Error", re.I), "RuntimeError", "runtime_error"),
-    (re.compile(r"Memory_?Error|OutOfMemory|OOM\b", re.I), "MemoryError",
-     "memory_error"),
+    (re.compile(r"Memory_?Error|OutOfMemory|OOM\b|bad_alloc|"
+                r"(memcpy|sizeof|memory).{0,20}(error|fail)", re.I),
+     "MemoryError", "memory_error"),
     (re.compile(r"TypeError", re.I), "TypeError", "type_error"),
     (re.compile(r"ImportError|ModuleNotFoundError", re.I), "ImportError",
      "import_error"),
     (re.compile(r"KeyError", re.I), "KeyError", "key_error"),
     (re.compile(r"AssertionError", re.I), "AssertionError", "AssertionError"),
-    (re.compile(r"File(NotFound)?Error|IOError|FileExistsError", re.I),
+    (re.compile(r"File(NotFound)?Error|IOError|FileExistsError|"
+                r"EndOfFileException|file.{0,12}(error|exception|fail)", re.I),
      "FileError", "FileError"),
     (re.compile(r"NotImplementedError", re.I), "NotImplementedError",
      "NotImplementedError"),
@@ -45,16 +47,22 @@ RE_RAISES = re.compile(
     r"assertRaises\w*|pytest\.raises|self\.raises|with raises|EXPECT_THROW|"
     r"EXPECT_ANY_THROW|ASSERT_THROW|ASSERT_DEATH|EXPECT_DEATH|expect\(.*\)\.to\.throw",
     re.I)
+# calibrated on the reference gold labels (classify/agreement.py):
+# EXPECT/ASSERT_NEAR + epsilon belong to absolute/relative tolerance there,
+# assertAlmostEqual/round to rounding, fabs/assertLess-style bounds to
+# error_bounding.
 RE_ALMOST = re.compile(
     r"assertAlmostEqual|assert_almost_equal|assertNotAlmostEqual|round\(|"
-    r"EXPECT_NEAR|ASSERT_NEAR|pytest\.approx|\bapprox\(|places\s*=|decimal\s*=", re.I)
+    r"pytest\.approx|\bapprox\(|places\s*=|decimal\s*=|rounding", re.I)
 RE_TOLERANCE = re.compile(
     r"assert_allclose|allclose|assert_array_almost_equal|atol|rtol|"
-    r"tolerance|abs_error|rel_error|relative|EXPECT_(FLOAT|DOUBLE)_EQ|"
-    r"isclose", re.I)
+    r"tolerance|abs_error|rel_error|relative error|EXPECT_(FLOAT|DOUBLE)_EQ|"
+    r"EXPECT_NEAR|ASSERT_NEAR|epsilon|isclose", re.I)
 RE_ERR_BOUND = re.compile(
-    r"(abs|fabs|np\.abs)\s*\(.+[-−].+\)\s*[<>]=?|error\w*\s*[<>]=?\s*|"
-    r"loss\w*\s*[<>]=?\s*|\bmse\b|\brmse\b", re.I)
+    r"(\b|_)(fabs|abs)\s*\(.+[-−].+\)\s*[<>]=?|error\w*\s*[<>]=?|"
+    r"loss\w*\s*[<>]=?|\bmse\b|\brmse\b|WithinEpsilon|error.?bound|"
+    r"assertLess\w*\(.*(err|loss|norm|prob|dist)|EXPECT_LT\(.*(err|norm)",
+    re.I)
 RE_INSTANCE = re.compile(
     r"isinstance|assertIsInstance|assertNotIsInstance|\btype\s*\(\s*[\w.\[\]]+\s*\)\s*(==|is)\b|"
     r"\.dtype\s*==|instanceof|dynamic_cast", re.I)
@@ -62,20 +70,23 @@ RE_SUBSET = re.compile(
     r"assert(Not)?In\b|\bin\s+(list|set|dict|keys|\w+\.keys)|issubset|"
     r"assertDictContainsSubset|\bcontains\b|EXPECT_TRUE\(.*find\(", re.I)
 RE_RANGE = re.compile(
-    r"assert(Greater|Less)(Equal)?|assertBetween|[^=!<>][<>]=?[^=<>]|"
+    r"assert(Greater|Less)(Equal)?|assertBetween|\brange\b|"
+    r"[<>]=?\s*-?\d|\d\s*[<>]=?|"
     r"EXPECT_[GL][ET]\b|ASSERT_[GL][ET]\b", re.I)
 RE_STATUS = re.compile(
     r"assert(True|False)\b|EXPECT_TRUE|EXPECT_FALSE|ASSERT_TRUE|ASSERT_FALSE|"
     r"\.ok\(\)|status|is_(alive|ready|running|done|finished|initialized)|"
     r"succe(ss|eded)|\bfailed\b|\.to\.be\.(true|false)", re.I)
 RE_LOGICAL = re.compile(
-    r"\b(and|or|not)\b|&&|\|\||assertLogicalExpression", 0)
+    r"logical ?(statement|condition|expression)?|\b(and|or)\b|&&|\|\|",
+    re.I)
 RE_NULL = re.compile(
     r"assertIs(Not)?None|is\s+(not\s+)?None|nullptr|!=\s*NULL|==\s*NULL|"
     r"\bNone\b\s*(==|!=|is)|EXPECT_EQ\(nullptr", re.I)
+# gold "negative_test" = asserting the negated condition (EXPECT_FALSE etc)
 RE_NEGATIVE = re.compile(
-    r"invalid|illegal|bad_|_bad\b|malformed|wrong|corrupt|negative_test|"
-    r"should_fail|fails?_on|with raises|assertRaises|pytest\.raises", re.I)
+    r"assertFalse|EXPECT_FALSE|ASSERT_FALSE|assertNot[A-Z]|assertIsNot\b|"
+    r"\bnot\b|negative_test|\.to\.be\.false", re.I)
 RE_EQUAL = re.compile(
     r"assert(Not)?Equals?\b|assert_equal|assertSequenceEqual|assertListEqual|"
     r"assertDictEqual|assertTupleEqual|assertCountEqual|==|!=|EXPECT_EQ|"
@@ -172,16 +183,13 @@ def classify_text(text: str, name: str = "", path: str = "") -> TestCaseRow:
     # --- oracle / strategy layer ---
     raises = RE_RAISES.search(t)
     err_type, err_strategy = "", None
-    if raises or re.search(r"raises|throw|exception|error", t, re.I):
-        for rx, etype, strat in ERROR_NAME_MAP:
-            if rx.search(t):
-                err_type, err_strategy = etype, strat
-                break
-    if raises:
+    for rx, etype, strat in ERROR_NAME_MAP:
+        if rx.search(t):
+            err_type, err_strategy = etype, strat
+            break
+    if raises or err_type:
         flags["error_handling"] = 1
-        if not err_type:
-            err_type = "Other_Error"
-        row.error_type = err_type
+        row.error_type = err_type or "Other_Error"
 
     if RE_ALMOST.search(t):
         flags["Approximation"] = 1
