@@ -17,6 +17,23 @@ bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
                      short* __restrict__ y, long n_elem, int D) {
   long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
   long stride = (long)gridDim.x * BG_BLOCK * 8;
+  if (stride % D == 0) {
+    // fixed column window: hoist the bias load and the 64-bit modulo
+    const int col = (int)(idx0 % D);
+    short8_t b8 = *(const short8_t*)(b + col);
+    float bb[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) bb[j] = bf16_to_f32(b8[j]);
+    for (long i = idx0; i < n_elem; i += stride) {
+      short8_t v = *(const short8_t*)(x + i);
+      short8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[j]) + bb[j]));
+      *(short8_t*)(y + i) = o;
+    }
+    return;
+  }
   for (long i = idx0; i < n_elem; i += stride) {
     short8_t v = *(const short8_t*)(x + i);
     int col = (int)(i % D);  // D % 8 == 0 so the packet stays in one row

@@ -354,6 +354,7 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
 #pragma unroll
     for (int rr = 0; rr < 16; ++rr) {
       int kvl = (rr & 3) + 8 * (rr >> 2) + 4 * half;
+      if (kv_base + kvl >= L) continue;  // tail kv-block: no OOB rows
       long off = bh_sq + (long)(kv_base + kvl) * L + q0 + col;
       int slot = rr >> 1;
       p_t[off] = p_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];

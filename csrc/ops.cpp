@@ -136,6 +136,8 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b) {
   TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
   auto y = torch::empty_like(x);
   int grid = (int)std::min<long>((n / 8 + 255) / 256, 2048);
+  long g0f = D / std::__gcd((long)D, 2048L);
+  if (g0f <= 2048) grid = (int)((grid + g0f - 1) / g0f * g0f);
   CHECK_HIP(bias_gelu_fwd_launch(x.data_ptr(), b.data_ptr(), y.data_ptr(), n,
                                  D, grid, cur_stream()));
   return y;
