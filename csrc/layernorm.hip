@@ -146,7 +146,9 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
 __global__ void __launch_bounds__(BLOCK)
 ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
               const short* __restrict__ gamma, const float* __restrict__ mean_in,
-              const float* __restrict__ rstd_in, short* __restrict__ dx,
+              const float* __restrict__ rstd_in,
+              const short* __restrict__ ds_extra,  // optional += into dx
+              short* __restrict__ dx,
               float* __restrict__ ws_dgamma, float* __restrict__ ws_dbeta,
               int N, int D) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -171,6 +173,7 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
        row += gridDim.x * WAVES_PER_BLOCK) {
     const short* dyr = dy + (long)row * D;
     const short* xr = x + (long)row * D;
+    const short* der = ds_extra ? ds_extra + (long)row * D : nullptr;
     short* dxr = dx + (long)row * D;
     const float mean = mean_in[row], rstd = rstd_in[row];
     if (pkts <= MAX_PKT_BWD && D == pkts * WAVE * 8) {
@@ -198,11 +201,15 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       for (int p = 0; p < MAX_PKT_BWD; ++p) {
         if (p >= pkts) break;
         int base = (p * WAVE + lane) * 8;
+        short8_t ev;
+        if (der) ev = *(const short8_t*)(der + base);
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int k = p * 8 + j;
-          o[j] = f32_to_bf16(rstd * (dyg[k] - s1 - xh[k] * s2));
+          float d_ = rstd * (dyg[k] - s1 - xh[k] * s2);
+          if (der) d_ += bf16_to_f32(ev[j]);
+          o[j] = f32_to_bf16(d_);
           dg_acc[k] += dyv[k] * xh[k];
           db_acc[k] += dyv[k];
         }
@@ -227,13 +234,17 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
         short8_t vd = *(const short8_t*)(dyr + i);
         short8_t vx = *(const short8_t*)(xr + i);
         short8_t g8 = *(const short8_t*)(gamma + i);
+        short8_t ev;
+        if (der) ev = *(const short8_t*)(der + i);
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float d = bf16_to_f32(vd[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
           float g = d * bf16_to_f32(g8[j]);
-          o[j] = f32_to_bf16(rstd * (g - s1 - h * s2));
+          float d_ = rstd * (g - s1 - h * s2);
+          if (der) d_ += bf16_to_f32(ev[j]);
+          o[j] = f32_to_bf16(d_);
           atomicAdd(&sg[i + j], d * h);
           atomicAdd(&sb[i + j], d);
         }
@@ -291,14 +302,15 @@ hipError_t ln_fwd_launch(const void* x, const void* res, const void* gamma,
 }
 
 hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
-                         const void* mean, const void* rstd, void* dx,
+                         const void* mean, const void* rstd,
+                         const void* ds_extra, void* dx,
                          void* ws_dgamma, void* ws_dbeta, int N, int D,
                          int grid, hipStream_t stream) {
   size_t shm = (size_t)D * 2 * sizeof(float);
   ln_bwd_kernel<<<grid, BLOCK, shm, stream>>>(
       (const short*)dy, (const short*)x, (const short*)gamma,
-      (const float*)mean, (const float*)rstd, (short*)dx,
-      (float*)ws_dgamma, (float*)ws_dbeta, N, D);
+      (const float*)mean, (const float*)rstd, (const short*)ds_extra,
+      (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);
   return hipGetLastError();
 }
 

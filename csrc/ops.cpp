@@ -15,8 +15,8 @@ hipError_t ln_fwd_launch(const void*, const void*, const void*, const void*,
                          void*, void*, void*, void*, int, int, float, int,
                          hipStream_t);
 hipError_t ln_bwd_launch(const void*, const void*, const void*, const void*,
-                         const void*, void*, void*, void*, int, int, int,
-                         hipStream_t);
+                         const void*, const void*, void*, void*, void*, int,
+                         int, int, hipStream_t);
 hipError_t colsum_launch(const void*, void*, void*, int, int, hipStream_t);
 hipError_t bias_gelu_fwd_launch(const void*, const void*, void*, long, int,
                                 int, hipStream_t);
@@ -106,7 +106,8 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x,
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor gamma, torch::Tensor mean,
-                                         torch::Tensor rstd) {
+                                         torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> ds_extra) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(gamma, "gamma");
   const int D = (int)x.size(-1);
   const long N = x.numel() / D;
@@ -115,8 +116,14 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto ws_dg = torch::empty({grid, D}, opts);
   auto ws_db = torch::empty({grid, D}, opts);
+  const void* de = nullptr;
+  if (ds_extra.has_value()) {
+    check_bf16(*ds_extra, "ds_extra");
+    TORCH_CHECK(ds_extra->numel() == x.numel(), "ds_extra shape mismatch");
+    de = ds_extra->data_ptr();
+  }
   CHECK_HIP(ln_bwd_launch(dy.data_ptr(), x.data_ptr(), gamma.data_ptr(),
-                          mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
+                          mean.data_ptr(), rstd.data_ptr(), de, dx.data_ptr(),
                           ws_dg.data_ptr(), ws_db.data_ptr(), (int)N, D, grid,
                           cur_stream()));
   auto dgamma = torch::empty({D}, opts);

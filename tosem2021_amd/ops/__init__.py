@@ -59,7 +59,8 @@ class _LayerNormFn(torch.autograd.Function):
         x, gamma, mean, rstd = ctx.saved_tensors
         dy = dy.contiguous()
         if x.is_cuda:
-            dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy, x, gamma, mean, rstd)
+            dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy, x, gamma, mean,
+                                                        rstd, None)
         else:
             dx, dgamma, dbeta = reference.layernorm_bwd(dy, x, gamma, mean, rstd)
         return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
@@ -89,11 +90,14 @@ class _AddLayerNormFn(torch.autograd.Function):
         s, gamma, mean, rstd = ctx.saved_tensors
         dy = dy.contiguous()
         if s.is_cuda:
-            dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy, s, gamma, mean, rstd)
+            # the downstream residual grad ds is added INSIDE the kernel
+            de = ds.contiguous() if ds is not None else None
+            dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy, s, gamma, mean,
+                                                        rstd, de)
         else:
             dx, dgamma, dbeta = reference.layernorm_bwd(dy, s, gamma, mean, rstd)
-        if ds is not None:
-            dx = dx + ds
+            if ds is not None:
+                dx = dx + ds
         return dx, dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
 
 
