@@ -279,3 +279,18 @@ def test_flash_bwd_ds_vs_reference():
     assert torch.allclose(ds_t.float().transpose(-1, -2), ds_ref, atol=2e-2,
                           rtol=3e-2), \
         (ds_t.float().transpose(-1, -2) - ds_ref).abs().max()
+
+
+def test_layernorm_bwd_with_residual_grad():
+    torch.manual_seed(15)
+    N, D = 64, 1024
+    xg = _bf16(torch.randn(N, D))
+    gg = _bf16(torch.randn(D))
+    bg = _bf16(torch.randn(D))
+    dyg = _bf16(torch.randn(N, D))
+    ds = _bf16(torch.randn(N, D))
+    _, _, mean, rstd = ops.hip_ops().layernorm_fwd(xg, None, gg, bg, 1e-5)
+    dx, _, _ = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd, ds)
+    dx0, _, _ = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd, None)
+    expect = (dx0.float() + ds.float()).to(torch.bfloat16)
+    assert torch.allclose(dx.float(), expect.float(), atol=2e-2, rtol=2e-2)
