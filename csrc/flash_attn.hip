@@ -173,7 +173,12 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
 #pragma unroll
       for (int r = 0; r < 16; ++r) o_acc[t][r] *= arow[r];
 
-    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap) ----
+    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
+    // The pack uses compiler-generated v_cvt_pk_bf16_f32 (via
+    // __float22bfloat162_rn) so hipcc's hazard recognizer pads the
+    // VALU-write -> v_permlane32_swap window itself (guide T21 hazard: an
+    // inline-asm cvt feeding the permlane builtin is under-padded by one
+    // wait state and breaks under register pressure).
     short8_t pf[2];
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
@@ -183,11 +188,10 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       for (int i = 0; i < 2; ++i) {
         int r0 = c * 8 + 2 * i;
         int r1 = c * 8 + 4 + 2 * i;
-        unsigned lo, hi;
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(lo)
-            : "v"(sv[r0]), "v"(sv[r0 + 1]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(hi)
-            : "v"(sv[r1]), "v"(sv[r1 + 1]));
+        __hip_bfloat162 plo = __float22bfloat162_rn({sv[r0], sv[r0 + 1]});
+        __hip_bfloat162 phi = __float22bfloat162_rn({sv[r1], sv[r1 + 1]});
+        unsigned lo = *reinterpret_cast<unsigned*>(&plo);
+        unsigned hi = *reinterpret_cast<unsigned*>(&phi);
         auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
         u[i] = sw[0];
         u[i + 2] = sw[1];
