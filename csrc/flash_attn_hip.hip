@@ -43,7 +43,7 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
@@ -165,14 +165,14 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     m_run = m_new;
 
     alpha_lds[wid * 32 + col] = a;
-    float arow[16];
 #pragma unroll
-    for (int r = 0; r < 16; ++r)
-      arow[r] = alpha_lds[wid * 32 + (r & 3) + 8 * (r >> 2) + 4 * half];
+    for (int t = 0; t < 2; ++t) {
 #pragma unroll
-    for (int t = 0; t < 2; ++t)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) o_acc[t][r] *= arow[r];
+      for (int r = 0; r < 16; ++r) {
+        int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        o_acc[t][r] *= alpha_lds[wid * 32 + qrow];
+      }
+    }
 
     // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
     // The pack uses compiler-generated v_cvt_pk_bf16_f32 (via
@@ -189,10 +189,10 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       for (int i = 0; i < 2; ++i) {
         int r0 = c * 8 + 2 * i;
         int r1 = c * 8 + 4 + 2 * i;
-        unsigned lo = __builtin_bit_cast(
-            unsigned, __float22bfloat162_rn({sv[r0], sv[r0 + 1]}));
-        unsigned hi = __builtin_bit_cast(
-            unsigned, __float22bfloat162_rn({sv[r1], sv[r1 + 1]}));
+        __hip_bfloat162 plo = __float22bfloat162_rn({sv[r0], sv[r0 + 1]});
+        __hip_bfloat162 phi = __float22bfloat162_rn({sv[r1], sv[r1 + 1]});
+        unsigned lo = *reinterpret_cast<unsigned*>(&plo);
+        unsigned hi = *reinterpret_cast<unsigned*>(&phi);
         auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
         u[i] = sw[0];
         u[i + 2] = sw[1];
@@ -218,18 +218,15 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   if (q_valid && lse != nullptr && half == 0)
     lse[(long)bh * L + my_q] = m_run + __logf(l_run);
   alpha_lds[wid * 32 + col] = 1.0f / l_run;
-  float ilrow[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r)
-    ilrow[r] = alpha_lds[wid * 32 + (r & 3) + 8 * (r >> 2) + 4 * half];
 #pragma unroll
   for (int t = 0; t < 2; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      int qrow = q_base + (r & 3) + 8 * (r >> 2) + 4 * half;
+      int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
+      int qrow = q_base + rloc;
       if (qrow >= L) continue;
       o[bh_off + (long)qrow * FA_DH + 32 * t + col] =
-          f32_to_bf16(o_acc[t][r] * ilrow[r]);
+          f32_to_bf16(o_acc[t][r] * alpha_lds[wid * 32 + rloc]);
     }
   }
 }

@@ -42,7 +42,7 @@ def test_layernorm_bwd(N, D):
     bg = _bf16(torch.randn(D))
     dyg = _bf16(torch.randn(N, D))
     _, _, mean, rstd = ops.hip_ops().layernorm_fwd(xg, None, gg, bg, 1e-5)
-    dx, dgamma, dbeta = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd)
+    dx, dgamma, dbeta = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd, None)
     dxe, dge, dbe = ref.layernorm_bwd(dyg.float().cpu(), xg.float().cpu(),
                                       gg.float().cpu(), mean.cpu(), rstd.cpu())
     assert torch.allclose(dx.float().cpu(), dxe, atol=4e-2, rtol=3e-2)
