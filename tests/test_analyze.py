@@ -88,3 +88,12 @@ def test_load_taxonomy_rejects_bad_schema(tmp_path):
     pd.DataFrame({"a": [1]}).to_csv(p, index=False)
     with pytest.raises(ValueError):
         load_taxonomy(str(p))
+
+
+def test_rq3_correlation_all_encodings():
+    df = _toy_df()
+    cnt = rq3_strategy_property_correlation(df, encode_cells="count")
+    assert cnt.loc["status_analysis", "Correctness"] == 1
+    ltx = rq3_strategy_property_correlation(df, encode_cells="latex")
+    cell = ltx.loc["status_analysis", "Correctness"]
+    assert cell.startswith("$Ray:") and "\\%$" in cell

@@ -84,15 +84,21 @@ def rq3_strategies_by_repo(df: pd.DataFrame) -> pd.DataFrame:
 
 
 def rq3_strategy_property_correlation(df: pd.DataFrame,
-                                      encode_cells: bool = True
+                                      encode_cells: str | bool = True
                                       ) -> pd.DataFrame:
-    """Strategy x property matrix.
+    """Strategy x property matrix in the reference's three cell encodings
+    (SURVEY.md §7 hard part 3):
 
-    encode_cells=True produces the reference's cell format
-    (tests_correlate_rq3.csv): 'repo:(x%), repo:(y%), ...' where x% is the
-    share of that repo's strategy-labeled rows carrying both labels; a cell
-    with no co-occurrence anywhere is the scalar 0.
+      encode_cells=True / "percent": 'repo:(x%), ...'  (tests_correlate_rq3.csv)
+      encode_cells="latex":          '$repo:x\%$, ...' (tests_correlate_rq4.csv)
+      encode_cells=False / "count":  integer co-occurrence counts
+                                     (tests_combined_correlate_rq3.csv)
+
+    x% = share of that repo's strategy-labeled rows carrying both labels;
+    a cell with no co-occurrence anywhere is the scalar 0.
     """
+    if encode_cells == "count":
+        encode_cells = False
     strat_sets = row_strategies(df)
     prop_sets = row_properties(df)
     repos = df["Repo"].astype(str)
@@ -113,7 +119,11 @@ def rq3_strategy_property_correlation(df: pd.DataFrame,
             if cell is None or not any(cell.values()):
                 out.loc[s, p] = 0 if encode_cells else 0.0
                 continue
-            if encode_cells:
+            if encode_cells == "latex":
+                out.loc[s, p] = "".join(
+                    f"${r}:{round(cell[r] / max(repo_tot[r], 1) * 100, 2)}\\%$, "
+                    for r in REPOS if cell[r])
+            elif encode_cells:
                 out.loc[s, p] = "".join(
                     f"{r}:({round(cell[r] / max(repo_tot[r], 1) * 100, 2)}%), "
                     for r in REPOS)
@@ -183,6 +193,15 @@ def write_all(df: pd.DataFrame, out_dir: str) -> Dict[str, str]:
     corr = rq3_strategy_property_correlation(df)
     p = os.path.join(out_dir, "RQ3", "tests_correlate_rq3.csv")
     corr.to_csv(p); paths["rq3_correlate"] = p
+    cnt = rq3_strategy_property_correlation(df, encode_cells="count")
+    p = os.path.join(out_dir, "RQ3", "tests_combined_correlate_rq3.csv")
+    cnt.to_csv(p); paths["rq3_correlate_counts"] = p
+    ltx = rq3_strategy_property_correlation(df, encode_cells="latex")
+    p = os.path.join(out_dir, "RQ3", "tests_correlate_rq4.csv")
+    ltx.to_csv(p); paths["rq3_correlate_latex"] = p
+    st_t = st.transpose()
+    p = os.path.join(out_dir, "RQ3", "tests_strategy_transpose_rq3.csv")
+    st_t.to_csv(p); paths["rq3_strategies_transpose"] = p
     m = rq4_test_methods(df)
     p = os.path.join(out_dir, "RQ4", "tests_methods.csv")
     m.to_csv(p, index=False); paths["rq4"] = p
