@@ -174,11 +174,9 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
 
     // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
-    // The pack uses compiler-generated v_cvt_pk_bf16_f32 (via
-    // __float22bfloat162_rn) so hipcc's hazard recognizer pads the
-    // VALU-write -> v_permlane32_swap window itself (guide T21 hazard: an
-    // inline-asm cvt feeding the permlane builtin is under-padded by one
-    // wait state and breaks under register pressure).
+    // The cvt_pk asm strings end with s_nop 1: the permlane32_swap hazard
+    // rule needs 2 wait states after a VALU write of either operand, and
+    // hipcc pads only one state after an asm block (guide §5.7 item 2, T21).
     short8_t pf[2];
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
@@ -188,10 +186,11 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       for (int i = 0; i < 2; ++i) {
         int r0 = c * 8 + 2 * i;
         int r1 = c * 8 + 4 + 2 * i;
-        __hip_bfloat162 plo = __float22bfloat162_rn({sv[r0], sv[r0 + 1]});
-        __hip_bfloat162 phi = __float22bfloat162_rn({sv[r1], sv[r1 + 1]});
-        unsigned lo = *reinterpret_cast<unsigned*>(&plo);
-        unsigned hi = *reinterpret_cast<unsigned*>(&phi);
+        unsigned lo, hi;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
+            : "v"(sv[r0]), "v"(sv[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
+            : "v"(sv[r1]), "v"(sv[r1 + 1]));
         auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
         u[i] = sw[0];
         u[i + 2] = sw[1];
