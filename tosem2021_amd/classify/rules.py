@@ -287,5 +287,9 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
         r.repo = repo
         r.file_id = file_id
         r.component = component or case.file_rel
+        # assertion rows inherit the enclosing case's workflow stage (an
+        # assertion's own text rarely carries stage cues)
+        r.category = head.category
+        r.category2 = head.category2
         rows.append(r)
     return rows
