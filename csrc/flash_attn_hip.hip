@@ -58,10 +58,11 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int wid = tid / WAVE;
   const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
   const int half = lane >> 5;
-  char* k_lds = smem + wid * (K_LDS_BYTES / FA_WAVES + VT_LDS_BYTES);
-  char* vt_lds = k_lds + K_LDS_BYTES / FA_WAVES;      // [64][40] rows 80 B
-  float* alpha_lds = (float*)(smem + FA_WAVES * (K_LDS_BYTES / FA_WAVES +
-                                                 VT_LDS_BYTES));
+  // each wave owns a FULL K tile (4 KB) + V^T tile (5 KB) slice; waves are
+  // free-running (no barriers), so slices must never overlap
+  char* k_lds = smem + wid * (K_LDS_BYTES + VT_LDS_BYTES);
+  char* vt_lds = k_lds + K_LDS_BYTES;                 // [64][40] rows 80 B
+  float* alpha_lds = (float*)(smem + FA_WAVES * (K_LDS_BYTES + VT_LDS_BYTES));
 
   const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
@@ -237,7 +238,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        hipStream_t stream) {
   int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   dim3 grid(B * H * n_qblocks);
-  size_t shm = FA_WAVES * (K_LDS_BYTES / FA_WAVES + VT_LDS_BYTES) +
+  size_t shm = FA_WAVES * (K_LDS_BYTES + VT_LDS_BYTES) +
                FA_WAVES * 32 * sizeof(float);
  hipLaunchKernelGGL(( flash_fwd_kernel), dim3(grid), dim3(FA_BLOCK), shm, stream, 
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
