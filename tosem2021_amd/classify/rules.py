@@ -161,10 +161,11 @@ STAGE_RULES = [
         r"train|\bfit\b|optimiz|gradient|loss|epoch|learn|model|network|"
         r"classif|regress|pipeline", re.I)),
     ("data_post", re.compile(
-        r"postprocess|predict|output|decode|nms|export_result|evaluat", re.I)),
+        r"postprocess|predict|output|decode|nms|export_result|evaluat|"
+        r"perception|detect|segment|track|fusion|camera|lidar|radar", re.I)),
     ("model_deployment", re.compile(
         r"deploy|serv(e|ing)|export|inference|onnx|tflite|compile_model|"
-        r"runtime", re.I)),
+        r"runtime|planning|control\b|routing|localization|canbus", re.I)),
     ("Monitoring", re.compile(r"monitor|metric|logg|dashboard|profil|trace", re.I)),
     ("config_utility", re.compile(
         r"config|flag|option|\butil|helper|param(s|eter)?\b|setting", re.I)),
