@@ -43,39 +43,38 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
                  int B, int H, int L, float scale) {
-  // v2: each wave stages its OWN K/V tile into a private LDS slice — no
-  // __syncthreads in the kv loop — and issues tile t+1's global loads before
-  // computing tile t (T14 split), so HBM/L2 latency hides under the MFMAs.
-  // Global K/V re-reads across q-blocks hit the XCD's L2 (xcd_group_remap).
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* vt_lds = (short*)(smem + K_LDS_BYTES);      // [64][40] (80 B rows)
+  float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
+
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
   const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
   const int half = lane >> 5;
-  // each wave owns a FULL K tile (4 KB) + V^T tile (5 KB) slice; waves are
-  // free-running (no barriers), so slices must never overlap
-  char* k_lds = smem + wid * (K_LDS_BYTES + VT_LDS_BYTES);
-  char* vt_lds = k_lds + K_LDS_BYTES;                 // [64][40] rows 80 B
-  float* alpha_lds = (float*)(smem + FA_WAVES * (K_LDS_BYTES + VT_LDS_BYTES));
 
+  // grid: x = (b*H + h) * n_qblocks + qb; remapped so one (b,h)'s q-blocks
+  // share an XCD and re-read K/V from that XCD's L2 (guide T1)
   const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
   int bh = bid / n_qblocks;
   int qb = bid % n_qblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
-  const int q_base = qb * FA_QWG + wid * FA_QB;
-  const int my_q = q_base + col;
+  const int q_base_wg = qb * FA_QWG;
+  const int q_base = q_base_wg + wid * FA_QB;        // this wave's 32 q rows
+  const int my_q = q_base + col;                      // this lane's q row
   const bool q_valid = my_q < L;
   const float* mrow = mask ? mask + (long)b * L : nullptr;
 
-  // Q fragments (once): lane holds Q[my_q][16c + 8*half + j]
+  // ---- load Q fragments (once): A/B-operand layout, 4 k-chunks of 16 ----
+  // lane holds Q[my_q][16c + 8*half + j], j=0..8
   short8_t qf[4];
   {
     const short* qr = q + bh_off + (long)(q_valid ? my_q : L - 1) * FA_DH;
@@ -84,64 +83,47 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       qf[c] = *(const short8_t*)(qr + c * 16 + half * 8);
   }
 
-  f32x16 o_acc[2];
+  f32x16 o_acc[2];                 // O[d-tile t][16 q-rows], d = 32t + col
 #pragma unroll
   for (int t = 0; t < 2; ++t) o_acc[t] = (f32x16)(0.f);
   float m_run = -3.0e38f;
   float l_run = 0.f;
 
-  // staging registers: this wave's slice of the K and V tiles.
-  // lane covers row pair {2*(lane>>4), +1} x 32-elem half (lane>>4 in 0..3):
-  // simpler: lane loads K[row = lane/2][32*(lane&1) .. +32) as 2 short8
-  const int srow = lane >> 1;            // 0..31
-  const int scol = (lane & 1) * 32;      // this lane: features scol..scol+31
-  short8_t kst[4], vst[4];
-  const int n_kv = L / FA_KVB;
-  {
-    const short* kr = k + bh_off + (long)srow * FA_DH + scol;
-    const short* vr = v + bh_off + (long)srow * FA_DH + scol;
-#pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      kst[p] = *(const short8_t*)(kr + 8 * p);
-      vst[p] = *(const short8_t*)(vr + 8 * p);
-    }
-  }
-
+  const int n_kv = L / FA_KVB;     // host asserts L % 32 == 0
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
-    // ---- write tile t from regs to this wave's LDS slice ----
-#pragma unroll
-    for (int piece = 0; piece < 4; ++piece) {
-      int cbyte = (scol + piece * 8) * 2;
-      *(short8_t*)(k_lds + srow * 128 + kswz(srow, cbyte)) = kst[piece];
+    // ---- stage K (swizzled) and V^T cooperatively ----
+    __syncthreads();
+    {
+      // K: 256 threads x one short8: row = tid/8 (32 rows), col8 = tid%8
+      int row = tid >> 3, c8 = (tid & 7) * 16;  // byte col
+      short8_t kv8 = *(const short8_t*)(k + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
+      *(short8_t*)((char*)k_lds + row * 128 + kswz(row, c8)) = kv8;
+      // V: same global packet, transposed scatter into vt_lds
+      short8_t vv8 = *(const short8_t*)(v + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int d = scol + piece * 8 + j;
-        *(short*)(vt_lds + d * VT_ROW_BYTES + srow * 2) = vst[piece][j];
+        int d = (c8 >> 1) + j;
+        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + row * 2) = vv8[j];
       }
     }
-    // ---- issue tile t+1 global loads (stay in flight under the MFMAs) ----
-    if (kt + 1 < n_kv) {
-      const short* kr = k + bh_off + (long)(kv0 + FA_KVB + srow) * FA_DH + scol;
-      const short* vr = v + bh_off + (long)(kv0 + FA_KVB + srow) * FA_DH + scol;
-#pragma unroll
-      for (int p = 0; p < 4; ++p) {
-        kst[p] = *(const short8_t*)(kr + 8 * p);
-        vst[p] = *(const short8_t*)(vr + 8 * p);
-      }
-    }
+    __syncthreads();
 
-    // ---- S^T = K @ Q^T ----
+    // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
     f32x16 s_acc = (f32x16)(0.f);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
+      // K fragment: lane holds K[kv=col][16c + 8*half + j] from swizzled LDS
+      int row = col;
       int byte_off = (16 * c + 8 * half) * 2;
-      short8_t kf = *(const short8_t*)(k_lds + col * 128 +
-                                       kswz(col, byte_off));
+      short8_t kf = *(const short8_t*)((char*)k_lds + row * 128 +
+                                       kswz(row, byte_off));
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc, 0, 0, 0);
     }
 
-    // ---- online softmax (lane-local q column) ----
+    // ---- online softmax (per lane: one q column, 16 kv rows) ----
     float sv[16];
     float tmax = -3.0e38f;
 #pragma unroll
@@ -165,6 +147,8 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     l_run = l_run * a + rsum;
     m_run = m_new;
 
+    // broadcast alpha(q) to the O accumulator rows via per-wave LDS
+    // (both halves write the same value; wave-internal ds ordering suffices)
     alpha_lds[wid * 32 + col] = a;
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
@@ -176,9 +160,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
 
     // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
-    // The cvt_pk asm strings end with s_nop 1: the permlane32_swap hazard
-    // rule needs 2 wait states after a VALU write of either operand, and
-    // hipcc pads only one state after an asm block (guide §5.7 item 2, T21).
+    // chunk 0: kv 8*half..+8 from regs 0..7; chunk 1: kv 16+8*half from 8..15
     short8_t pf[2];
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
@@ -186,35 +168,39 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       uint4_t u;
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
-        int r0 = c * 8 + 2 * i;
-        int r1 = c * 8 + 4 + 2 * i;
+        int r0 = c * 8 + 2 * i;         // own pair (kv low quad)
+        int r1 = c * 8 + 4 + 2 * i;     // pair the partner half needs (+8)
         unsigned lo, hi;
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(lo)
             : "v"(sv[r0]), "v"(sv[r0 + 1]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(hi)
             : "v"(sv[r1]), "v"(sv[r1 + 1]));
         auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
+        // consecutive-kv order: {01, 23, 45, 67} (own pairs i=0,1 first)
         u[i] = sw[0];
         u[i + 2] = sw[1];
       }
       pf[c] = __builtin_bit_cast(short8_t, u);
     }
 
-    // ---- PV: O += P^T @ V ----
+    // ---- PV: O[q, d] += P^T @ V — B-fragment from transposed V tile ----
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
+        // lane holds V[kv = 16c + 8*half + j][d = 32t + col]
         int d = 32 * t + col;
         int kvb = 16 * c + 8 * half;
-        short8_t vf = *(const short8_t*)(vt_lds + d * VT_ROW_BYTES + kvb * 2);
+        short8_t vf = *(const short8_t*)((char*)vt_lds + d * VT_ROW_BYTES +
+                                         kvb * 2);
         o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
                                                            o_acc[t], 0, 0, 0);
       }
     }
   }
 
-  // ---- epilogue ----
+  // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
+  //      store O rows + logsumexp ----
   if (q_valid && lse != nullptr && half == 0)
     lse[(long)bh * L + my_q] = m_run + __logf(l_run);
   alpha_lds[wid * 32 + col] = 1.0f / l_run;
@@ -238,8 +224,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        hipStream_t stream) {
   int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   dim3 grid(B * H * n_qblocks);
-  size_t shm = FA_WAVES * (K_LDS_BYTES + VT_LDS_BYTES) +
-               FA_WAVES * 32 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
  hipLaunchKernelGGL(( flash_fwd_kernel), dim3(grid), dim3(FA_BLOCK), shm, stream, 
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale);
