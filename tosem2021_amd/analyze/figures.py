@@ -4,7 +4,7 @@ from __future__ import annotations
 
 import html
 import os
-from typing import Dict, List, Sequence
+from typing import Dict, Sequence
 
 import pandas as pd
 

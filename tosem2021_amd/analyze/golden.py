@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import math
 import os
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import pandas as pd
 

@@ -14,7 +14,7 @@ are validated in tests/test_analyze_golden.py against the shipped CSVs.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import pandas as pd
 

@@ -6,7 +6,7 @@ analysis layer (rq1/rq3/rq4) consumes the normalized form produced here.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Set
+from typing import Dict, List, Set
 
 import pandas as pd
 

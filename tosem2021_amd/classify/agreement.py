@@ -16,7 +16,7 @@ import pandas as pd
 from tosem2021_amd.analyze.taxonomy import (
     row_method, row_properties, row_strategies)
 from tosem2021_amd.classify.rules import classify_text
-from tosem2021_amd.extract.schema import METHODS, PROPERTIES, STRATEGIES
+from tosem2021_amd.extract.schema import PROPERTIES, STRATEGIES
 
 
 @dataclass

@@ -6,8 +6,6 @@ The one place the reference's capability legitimately meets the GPU
 """
 from __future__ import annotations
 
-import math
-import os
 import time
 from typing import Dict, Optional
 

@@ -15,11 +15,10 @@ codebooks were partially stripped from the reference:
 from __future__ import annotations
 
 import re
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
-from tosem2021_amd.extract.python_extractor import Assertion, TestCase
-from tosem2021_amd.extract.schema import (STAGES, STRATEGIES, TestCaseRow)
+from tosem2021_amd.extract.python_extractor import TestCase
+from tosem2021_amd.extract.schema import TestCaseRow
 
 # ---------------------------------------------------------------------------
 # error-name -> (Error_Type value, strategy)

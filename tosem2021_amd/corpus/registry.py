@@ -6,7 +6,7 @@ the corpus root is configurable so the pipeline mines any checkout.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 

@@ -9,7 +9,7 @@ next round's table — plus the paper-style defaults.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 import pandas as pd

@@ -2,9 +2,8 @@
 data (tokens + multi-hot strategy/property labels + stage/method classes)."""
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
-from typing import Dict, Iterator, List, Optional, Tuple
+from typing import Dict, Iterator, List, Tuple
 
 import pandas as pd
 import torch

@@ -8,11 +8,8 @@ as the study's labelers did for e.g. auto-sklearn: taxonomy_test2.csv:2-5).
 from __future__ import annotations
 
 import ast
-import io
-import os
-import tokenize
 from dataclasses import dataclass, field
-from typing import Iterator, List, Optional
+from typing import List, Optional
 
 
 @dataclass

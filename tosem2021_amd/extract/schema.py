@@ -9,7 +9,7 @@ Mirrors the reference's master dataset RQs/taxonomy_test2.csv:1 (41 columns,
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field, fields
+from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
 # The reference master-CSV column order (taxonomy_test2.csv:1).

@@ -8,7 +8,7 @@ processes and machines (no PYTHONHASHSEED dependence).
 from __future__ import annotations
 
 import re
-from typing import List, Optional
+from typing import List
 
 import torch
 

@@ -4,12 +4,11 @@ from __future__ import annotations
 
 import csv
 import os
-from dataclasses import dataclass
-from typing import Dict, List, Optional, Sequence
+from typing import List, Optional, Sequence
 
 from tosem2021_amd.classify.rules import classify_case
 from tosem2021_amd.corpus import walker
-from tosem2021_amd.corpus.registry import PROJECTS, Project, project_root
+from tosem2021_amd.corpus.registry import PROJECTS, project_root
 from tosem2021_amd.extract.python_extractor import extract_file
 from tosem2021_amd.extract.schema import TAXONOMY_COLUMNS, TestCaseRow
 from tosem2021_amd.utils.metrics import get_metrics

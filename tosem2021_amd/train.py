@@ -15,15 +15,14 @@ from __future__ import annotations
 
 import math
 import os
-import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist
 
 from tosem2021_amd import ops
-from tosem2021_amd.models import MLTC, MLTCConfig, build_model
+from tosem2021_amd.models import MLTCConfig, build_model
 from tosem2021_amd.parallel.ddp import BucketedAllReduce
 from tosem2021_amd.utils.metrics import get_metrics
 from tosem2021_amd.utils.trace import trace

@@ -12,7 +12,7 @@ import os
 import threading
 import time
 from contextlib import contextmanager
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 
 class Tracer:
