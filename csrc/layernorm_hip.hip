@@ -161,12 +161,14 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   __syncthreads();
 
   const int pkts = D / (WAVE * 8);
-  // bwd caches 3 f32 arrays per lane -> cap at 4 packets (D <= 2048).
+  // bwd register-cached path capped at 2 packets (D <= 1024, the flagship
+  // width); wider D takes the re-read path. 4 packets put the kernel at 256
+  // VGPR = 1-2 waves/SIMD.
   // Each (wave,lane,j) owns a FIXED column set across its whole row loop, so
   // dgamma/dbeta partials accumulate in registers and hit the LDS once per
   // wave at the end (the per-element LDS atomics were 8.5 ms/step in the
   // baseline profile — profiles/r01_kernel_stats_baseline.md).
-#define MAX_PKT_BWD 4
+#define MAX_PKT_BWD 2
   float dg_acc[MAX_PKT_BWD * 8] = {0.f};
   float db_acc[MAX_PKT_BWD * 8] = {0.f};
   const bool cached = pkts <= MAX_PKT_BWD && D == pkts * WAVE * 8;
@@ -178,7 +180,10 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
     short* dxr = dx + (long)row * D;
     const float mean = mean_in[row], rstd = rstd_in[row];
     if (pkts <= MAX_PKT_BWD && D == pkts * WAVE * 8) {
-      float xh[MAX_PKT_BWD * 8], dyg[MAX_PKT_BWD * 8], dyv[MAX_PKT_BWD * 8];
+      // two register arrays only (xh, dyg); dgamma/dbeta accumulate in the
+      // first pass (the third array pushed the kernel to 256 VGPR = 1
+      // wave/SIMD)
+      float xh[MAX_PKT_BWD * 8], dyg[MAX_PKT_BWD * 8];
       float s1 = 0.f, s2 = 0.f;
 #pragma unroll
       for (int p = 0; p < MAX_PKT_BWD; ++p) {
@@ -189,10 +194,13 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
         short8_t g8 = *(const short8_t*)(gamma + base);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
+          int k = p * 8 + j;
           float d = bf16_to_f32(vd[j]);
           float h = (bf16_to_f32(vx[j]) - mean) * rstd;
           float g = d * bf16_to_f32(g8[j]);
-          xh[p * 8 + j] = h; dyg[p * 8 + j] = g; dyv[p * 8 + j] = d;
+          xh[k] = h; dyg[k] = g;
+          dg_acc[k] += d * h;
+          db_acc[k] += d;
           s1 += g; s2 += g * h;
         }
       }
@@ -211,8 +219,6 @@ ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
           float d_ = rstd * (dyg[k] - s1 - xh[k] * s2);
           if (der) d_ += bf16_to_f32(ev[j]);
           o[j] = f32_to_bf16(d_);
-          dg_acc[k] += dyv[k] * xh[k];
-          db_acc[k] += dyv[k];
         }
         *(short8_t*)(dxr + base) = o;
       }
