@@ -63,7 +63,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                      steps: int = 500, batch: int = 32, seq: int = 256,
                      lr: float = 3e-4, ckpt_dir: Optional[str] = None,
                      resume: bool = False, device: Optional[str] = None,
-                     eval_every: int = 0, seed: int = 0) -> dict:
+                     eval_every: int = 0, seed: int = 0,
+                     dropout: float = 0.0) -> dict:
     dev = torch.device(device) if device else (
         torch.device("cuda") if torch.cuda.is_available() else
         torch.device("cpu"))
@@ -74,7 +75,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
         full = TaxonomyDataset.from_taxonomy(df)
 
     base = CONFIGS[model]
-    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
+    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq,
+                     "dropout": dropout})
     if full._tokens is not None and int(full._tokens.max()) >= cfg.vocab_size:
         # prepared file was tokenized for a bigger vocab (e.g. mltc-base);
         # fold ids into this config's hash space deterministically
@@ -150,7 +152,8 @@ def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
         torch.device("cpu"))
     df = load_taxonomy(taxonomy_path)
     base = CONFIGS[model]
-    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
+    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq,
+                     "dropout": dropout})
     tok = CodeTokenizer(cfg.vocab_size)
     tcfg = TrainConfig(model=model, ckpt_dir=ckpt_dir,
                        dtype="bf16" if dev.type == "cuda" else "f32")

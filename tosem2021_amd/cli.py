@@ -78,7 +78,7 @@ def cmd_train(args):
     res = train_classifier(
         taxonomy_path=args.taxonomy, model=args.model, steps=args.steps,
         batch=args.batch, seq=args.seq, lr=args.lr, ckpt_dir=args.ckpt_dir,
-        resume=args.resume)
+        resume=args.resume, dropout=args.dropout)
     print(json.dumps(res, indent=2))
 
 
@@ -133,6 +133,7 @@ def main(argv=None):
     p.add_argument("--lr", type=float, default=3e-4)
     p.add_argument("--ckpt-dir", default=None)
     p.add_argument("--resume", action="store_true")
+    p.add_argument("--dropout", type=float, default=0.0)
     p.set_defaults(fn=cmd_train)
 
     args = ap.parse_args(argv)

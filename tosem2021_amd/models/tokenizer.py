@@ -61,7 +61,7 @@ class CodeTokenizer:
         """Returns (tokens [B, L] int64, mask [B, L] bool)."""
         rows = [self.encode(t, max_len) for t in texts]
         L = max(max(len(r) for r in rows), 1) if rows else 1
-        L = (L + 31) // 32 * 32  # flash kernel wants L % 32 == 0
+        L = (L + 63) // 64 * 64  # flash kernel stages 64-key tiles
         toks = torch.full((len(rows), L), PAD, dtype=torch.long)
         mask = torch.zeros(len(rows), L, dtype=torch.bool)
         for i, r in enumerate(rows):
