@@ -61,7 +61,9 @@ class CodeTokenizer:
         """Returns (tokens [B, L] int64, mask [B, L] bool)."""
         rows = [self.encode(t, max_len) for t in texts]
         L = max(max(len(r) for r in rows), 1) if rows else 1
-        L = (L + 63) // 64 * 64  # flash kernel stages 64-key tiles
+        # pad to 64 (flash kernel granularity) when it fits max_len, else 8
+        L64 = (L + 63) // 64 * 64
+        L = L64 if L64 <= max_len else (L + 7) // 8 * 8
         toks = torch.full((len(rows), L), PAD, dtype=torch.long)
         mask = torch.zeros(len(rows), L, dtype=torch.bool)
         for i, r in enumerate(rows):
