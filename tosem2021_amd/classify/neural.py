@@ -152,8 +152,7 @@ def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
         torch.device("cpu"))
     df = load_taxonomy(taxonomy_path)
     base = CONFIGS[model]
-    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq,
-                     "dropout": dropout})
+    cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq})
     tok = CodeTokenizer(cfg.vocab_size)
     tcfg = TrainConfig(model=model, ckpt_dir=ckpt_dir,
                        dtype="bf16" if dev.type == "cuda" else "f32")
