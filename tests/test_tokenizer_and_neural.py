@@ -68,3 +68,14 @@ def test_train_classifier_tiny_cpu(tmp_path):
         assert k in res
     # checkpoint written
     assert any(f.startswith("ckpt_") for f in os.listdir(tmp_path / "ck"))
+
+
+def test_linear_baseline_tiny(tmp_path):
+    from tosem2021_amd.classify.baseline import train_linear_baseline
+    tax = str(tmp_path / "tiny_tax.csv")
+    _write_tiny_taxonomy(tax, n=160)
+    res = train_linear_baseline(tax, dim=256, val_frac=0.25)
+    assert res["n_val"] == 40
+    # two perfectly-separable synthetic classes: the linear model must learn
+    assert res["strategy_micro_f1"] > 0.9, res
+    assert res["method_accuracy"] > 0.9
