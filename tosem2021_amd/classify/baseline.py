@@ -65,10 +65,15 @@ def train_linear_baseline(taxonomy_path: str, dim: int = 4096,
     props = fit_multilabel(train_ds.property_.numpy().astype(int),
                            val_ds.property_.numpy().astype(int), PROPERTIES)
 
-    from sklearn.linear_model import LogisticRegression as LR
-    mclf = LR(max_iter=max_iter, C=4.0)
-    mclf.fit(Xtr, train_ds.method.numpy())
-    macc = float((mclf.predict(Xva) == val_ds.method.numpy()).mean())
+    y_m = train_ds.method.numpy()
+    if len(np.unique(y_m)) < 2:
+        pred_m = np.full(len(val_ds), y_m[0] if len(y_m) else 0)
+    else:
+        from sklearn.linear_model import LogisticRegression as LR
+        mclf = LR(max_iter=max_iter, C=4.0)
+        mclf.fit(Xtr, y_m)
+        pred_m = mclf.predict(Xva)
+    macc = float((pred_m == val_ds.method.numpy()).mean())
 
     return {
         "n_train": len(train_ds),
