@@ -47,31 +47,33 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
                  int B, int H, int L, float scale) {
-  // Shared-tile structure, 64-key staging granularity: the workgroup stages
-  // K[64][64] (XOR-swizzled) + V^T[64][64] (transposed, 144-B rows) once per
-  // TWO 32-kv compute subtiles, halving barrier+staging overhead per MFMA.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                                  // 8 KiB swizzled
-  char* vt_lds = smem + 64 * 128;                      // [64 d][144 B rows]
-  float* alpha_lds = (float*)(smem + 64 * 128 + FA_DH * 144);
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* vt_lds = (short*)(smem + K_LDS_BYTES);      // [64][40] (80 B rows)
+  float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
-  const int col = lane & 31;
+  const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
   const int half = lane >> 5;
 
+  // grid: x = (b*H + h) * n_qblocks + qb; remapped so one (b,h)'s q-blocks
+  // share an XCD and re-read K/V from that XCD's L2 (guide T1)
   const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
   int bh = bid / n_qblocks;
   int qb = bid % n_qblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
-  const int q_base = qb * FA_QWG + wid * FA_QB;
-  const int my_q = q_base + col;
+  const int q_base_wg = qb * FA_QWG;
+  const int q_base = q_base_wg + wid * FA_QB;        // this wave's 32 q rows
+  const int my_q = q_base + col;                      // this lane's q row
   const bool q_valid = my_q < L;
   const float* mrow = mask ? mask + (long)b * L : nullptr;
 
+  // ---- load Q fragments (once): A/B-operand layout, 4 k-chunks of 16 ----
+  // lane holds Q[my_q][16c + 8*half + j], j=0..8
   short8_t qf[4];
   {
     const short* qr = q + bh_off + (long)(q_valid ? my_q : L - 1) * FA_DH;
@@ -80,119 +82,124 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       qf[c] = *(const short8_t*)(qr + c * 16 + half * 8);
   }
 
-  f32x16 o_acc[2];
+  f32x16 o_acc[2];                 // O[d-tile t][16 q-rows], d = 32t + col
 #pragma unroll
   for (int t = 0; t < 2; ++t) o_acc[t] = (f32x16)(0.f);
   float m_run = -3.0e38f;
   float l_run = 0.f;
 
-  const int n_big = L / 64;        // launcher asserts L % 64 == 0 for this path
-  for (int bt = 0; bt < n_big; ++bt) {
-    const int kv_big = bt * 64;
+  const int n_kv = L / FA_KVB;     // host asserts L % 32 == 0
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int kv0 = kt * FA_KVB;
+    // ---- stage K (swizzled) and V^T cooperatively ----
     __syncthreads();
     {
-      // K: thread covers 32 B of one row: row = tid/4, byte col = (tid&3)*32
-      int row = tid >> 2, c8 = (tid & 3) * 32;
-      const short* kr = k + bh_off + (long)(kv_big + row) * FA_DH + (c8 >> 1);
-      *(short8_t*)(k_lds + row * 128 + kswz(row, c8)) =
-          *(const short8_t*)(kr);
-      *(short8_t*)(k_lds + row * 128 + kswz(row, c8 + 16)) =
-          *(const short8_t*)(kr + 8);
-      const short* vr = v + bh_off + (long)(kv_big + row) * FA_DH + (c8 >> 1);
-      short8_t v0 = *(const short8_t*)(vr);
-      short8_t v1 = *(const short8_t*)(vr + 8);
+      // K: 256 threads x one short8: row = tid/8 (32 rows), col8 = tid%8
+      int row = tid >> 3, c8 = (tid & 7) * 16;  // byte col
+      short8_t kv8 = *(const short8_t*)(k + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
+      *(short8_t*)((char*)k_lds + row * 128 + kswz(row, c8)) = kv8;
+      // V: same global packet, transposed scatter into vt_lds
+      short8_t vv8 = *(const short8_t*)(v + bh_off + (long)(kv0 + row) * FA_DH +
+                                        (c8 >> 1));
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int d0 = (c8 >> 1) + j;
-        *(short*)(vt_lds + d0 * 144 + row * 2) = v0[j];
-        *(short*)(vt_lds + (d0 + 8) * 144 + row * 2) = v1[j];
+        int d = (c8 >> 1) + j;
+        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + row * 2) = vv8[j];
       }
     }
     __syncthreads();
 
+    // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
+    f32x16 s_acc = (f32x16)(0.f);
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-      const int kv0 = kv_big + 32 * sub;
-      f32x16 s_acc = (f32x16)(0.f);
-#pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        int row = 32 * sub + col;
-        int byte_off = (16 * c + 8 * half) * 2;
-        short8_t kf = *(const short8_t*)(k_lds + row * 128 +
-                                         kswz(row, byte_off));
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc,
-                                                        0, 0, 0);
-      }
+    for (int c = 0; c < 4; ++c) {
+      // K fragment: lane holds K[kv=col][16c + 8*half + j] from swizzled LDS
+      int row = col;
+      int byte_off = (16 * c + 8 * half) * 2;
+      short8_t kf = *(const short8_t*)((char*)k_lds + row * 128 +
+                                       kswz(row, byte_off));
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc, 0, 0, 0);
+    }
 
-      float sv[16];
-      float tmax = -3.0e38f;
+    // ---- online softmax (per lane: one q column, 16 kv rows) ----
+    float sv[16];
+    float tmax = -3.0e38f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+      float x = s_acc[r] * scale;
+      if (mrow) x += mrow[kv0 + kv_local];
+      sv[r] = x;
+      tmax = fmaxf(tmax, x);
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+    float m_new = fmaxf(m_run, tmax);
+    float a = __expf(m_run - m_new);
+    float rsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      sv[r] = __expf(sv[r] - m_new);
+      rsum += sv[r];
+    }
+    rsum += __shfl_xor(rsum, 32, WAVE);
+    l_run = l_run * a + rsum;
+    m_run = m_new;
+
+    // broadcast alpha(q) to the O accumulator rows via per-wave LDS
+    // (both halves write the same value; wave-internal ds ordering suffices)
+    alpha_lds[wid * 32 + col] = a;
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
-        float x = s_acc[r] * scale;
-        if (mrow) x += mrow[kv0 + kv_local];
-        sv[r] = x;
-        tmax = fmaxf(tmax, x);
+        int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        o_acc[t][r] *= alpha_lds[wid * 32 + qrow];
       }
-      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
-      float m_new = fmaxf(m_run, tmax);
-      float a = __expf(m_run - m_new);
-      float rsum = 0.f;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        sv[r] = __expf(sv[r] - m_new);
-        rsum += sv[r];
-      }
-      rsum += __shfl_xor(rsum, 32, WAVE);
-      l_run = l_run * a + rsum;
-      m_run = m_new;
+    }
 
-      alpha_lds[wid * 32 + col] = a;
+    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
+    // chunk 0: kv 8*half..+8 from regs 0..7; chunk 1: kv 16+8*half from 8..15
+    short8_t pf[2];
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
+    for (int c = 0; c < 2; ++c) {
+      typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+      uint4_t u;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-          o_acc[t][r] *= alpha_lds[wid * 32 + qrow];
-        }
+      for (int i = 0; i < 2; ++i) {
+        int r0 = c * 8 + 2 * i;         // own pair (kv low quad)
+        int r1 = c * 8 + 4 + 2 * i;     // pair the partner half needs (+8)
+        unsigned lo, hi;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
+            : "v"(sv[r0]), "v"(sv[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
+            : "v"(sv[r1]), "v"(sv[r1 + 1]));
+        auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
+        // consecutive-kv order: {01, 23, 45, 67} (own pairs i=0,1 first)
+        u[i] = sw[0];
+        u[i + 2] = sw[1];
       }
+      pf[c] = __builtin_bit_cast(short8_t, u);
+    }
 
-      // P -> bf16 A-fragments; asm cvt_pk ends with s_nop 1 (permlane hazard)
-      short8_t pf[2];
+    // ---- PV: O[q, d] += P^T @ V — B-fragment from transposed V tile ----
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
-        uint4_t u;
-#pragma unroll
-        for (int i = 0; i < 2; ++i) {
-          int r0 = c * 8 + 2 * i;
-          int r1 = c * 8 + 4 + 2 * i;
-          unsigned lo, hi;
-          asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
-              : "v"(sv[r0]), "v"(sv[r0 + 1]));
-          asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
-              : "v"(sv[r1]), "v"(sv[r1 + 1]));
-          auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
-          u[i] = sw[0];
-          u[i + 2] = sw[1];
-        }
-        pf[c] = __builtin_bit_cast(short8_t, u);
-      }
-
-#pragma unroll
-      for (int t = 0; t < 2; ++t) {
-#pragma unroll
-        for (int c = 0; c < 2; ++c) {
-          int d = 32 * t + col;
-          int kvb = 32 * sub + 16 * c + 8 * half;
-          short8_t vf = *(const short8_t*)(vt_lds + d * 144 + kvb * 2);
-          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
-                                                             o_acc[t], 0, 0, 0);
-        }
+        // lane holds V[kv = 16c + 8*half + j][d = 32t + col]
+        int d = 32 * t + col;
+        int kvb = 16 * c + 8 * half;
+        short8_t vf = *(const short8_t*)((char*)vt_lds + d * VT_ROW_BYTES +
+                                         kvb * 2);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
+                                                           o_acc[t], 0, 0, 0);
       }
     }
   }
 
+  // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
+  //      store O rows + logsumexp ----
   if (q_valid && lse != nullptr && half == 0)
     lse[(long)bh * L + my_q] = m_run + __logf(l_run);
   alpha_lds[wid * 32 + col] = 1.0f / l_run;
@@ -216,7 +223,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        hipStream_t stream) {
   int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   dim3 grid(B * H * n_qblocks);
-  size_t shm = 64 * 128 + FA_DH * 144 + FA_WAVES * 32 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale);

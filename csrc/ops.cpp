@@ -296,7 +296,7 @@ std::vector<torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(q.dim() == 4, "q must be [B, H, L, dh]");
   const long B = q.size(0), H = q.size(1), L = q.size(2), dh = q.size(3);
   TORCH_CHECK(dh == 64, "flash_fwd is specialized for head_dim 64");
-  TORCH_CHECK(L % 64 == 0, "flash_fwd needs L % 64 == 0");
+  TORCH_CHECK(L % 32 == 0, "flash_fwd needs L % 32 == 0");
   TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
   const void* mptr = nullptr;
   if (mask.has_value()) {

@@ -265,7 +265,7 @@ def flash_attention(q, k, v, mask=None, scale: float = 1.0):
 
 
 def flash_supported(head_dim: int, L: int) -> bool:
-    return head_dim == 64 and L % 64 == 0
+    return head_dim == 64 and L % 32 == 0
 
 
 def adamw_step(p, grad, m, v, master, *, lr, beta1=0.9, beta2=0.999, eps=1e-8,
