@@ -89,27 +89,34 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   float m_run = -3.0e38f;
   float l_run = 0.f;
 
+  // T14 async-stage split: tile t+1's global loads are issued before tile
+  // t's compute and land in registers while the MFMAs run; the LDS write of
+  // tile t happens between the barrier pair (write-late, guide G15).
   const int n_kv = L / FA_KVB;     // host asserts L % 32 == 0
+  const int srow = tid >> 3, sc8 = (tid & 7) * 16;  // this thread's 16-B piece
+  short8_t kv8 = *(const short8_t*)(k + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
+  short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
-    // ---- stage K (swizzled) and V^T cooperatively ----
-    __syncthreads();
+    __syncthreads();   // previous tile's LDS reads complete
     {
-      // K: 256 threads x one short8: row = tid/8 (32 rows), col8 = tid%8
-      int row = tid >> 3, c8 = (tid & 7) * 16;  // byte col
-      short8_t kv8 = *(const short8_t*)(k + bh_off + (long)(kv0 + row) * FA_DH +
-                                        (c8 >> 1));
-      *(short8_t*)((char*)k_lds + row * 128 + kswz(row, c8)) = kv8;
-      // V: same global packet, transposed scatter into vt_lds
-      short8_t vv8 = *(const short8_t*)(v + bh_off + (long)(kv0 + row) * FA_DH +
-                                        (c8 >> 1));
+      *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int d = (c8 >> 1) + j;
-        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + row * 2) = vv8[j];
+        int d = (sc8 >> 1) + j;
+        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + srow * 2) = vv8[j];
       }
     }
-    __syncthreads();
+    __syncthreads();   // tile t visible to all waves
+    if (kt + 1 < n_kv) {
+      // issue tile t+1 loads now; they stay in flight under the MFMAs
+      kv8 = *(const short8_t*)(k + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+    }
 
     // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
     f32x16 s_acc = (f32x16)(0.f);
@@ -300,20 +307,25 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
   }
 
   const int n_q = L / FA_KVB;
+  const int srow = tid >> 3, sc8 = (tid & 7) * 16;
+  short8_t qv8 = *(const short8_t*)(q + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
+  short8_t dv8 = *(const short8_t*)(dout + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
   for (int qt = 0; qt < n_q; ++qt) {
     const int q0 = qt * 32;
     __syncthreads();
     {
-      // stage Q and dO q-tiles, row-major XOR-swizzled (as fwd stages K)
-      int row = tid >> 3, c8 = (tid & 7) * 16;
-      short8_t qv8 = *(const short8_t*)(q + bh_off + (long)(q0 + row) * FA_DH +
-                                        (c8 >> 1));
-      *(short8_t*)((char*)q_lds + row * 128 + kswz(row, c8)) = qv8;
-      short8_t dv8 = *(const short8_t*)(dout + bh_off +
-                                        (long)(q0 + row) * FA_DH + (c8 >> 1));
-      *(short8_t*)((char*)do_lds + row * 128 + kswz(row, c8)) = dv8;
+      *(short8_t*)((char*)q_lds + srow * 128 + kswz(srow, sc8)) = qv8;
+      *(short8_t*)((char*)do_lds + srow * 128 + kswz(srow, sc8)) = dv8;
     }
     __syncthreads();
+    if (qt + 1 < n_q) {
+      qv8 = *(const short8_t*)(q + bh_off + (long)(q0 + 32 + srow) * FA_DH +
+                               (sc8 >> 1));
+      dv8 = *(const short8_t*)(dout + bh_off + (long)(q0 + 32 + srow) * FA_DH +
+                               (sc8 >> 1));
+    }
 
     // S^T = K @ Q^T and dP^T = V @ dO^T (B-fragments from the staged tiles:
     // lane reads row q=col, feature chunk 16c + 8*half)

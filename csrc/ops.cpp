@@ -44,6 +44,10 @@ hipError_t flash_bwd_ds_launch(const void*, const void*, const void*,
                                const void*, void*, void*, int, int, int,
                                float, hipStream_t);
 hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
+hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
+                                  int, int, hipStream_t);
+hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
+                                  void*, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -367,7 +371,43 @@ std::vector<torch::Tensor> flash_bwd_ds(torch::Tensor q, torch::Tensor k,
   return {p_t, ds_t};
 }
 
+std::vector<torch::Tensor> masked_pool_fwd(torch::Tensor x,
+                                           c10::optional<torch::Tensor> mask) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 3, "x must be [B, L, D]");
+  long B = x.size(0), L = x.size(1), D = x.size(2);
+  TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    TORCH_CHECK(mask->scalar_type() == torch::kBool && mask->is_contiguous(),
+                "mask must be contiguous bool");
+    mptr = mask->data_ptr();
+  }
+  auto pooled = torch::empty({B, D}, x.options());
+  auto counts = torch::empty({B}, x.options().dtype(torch::kFloat32));
+  CHECK_HIP(masked_pool_fwd_launch(x.data_ptr(), mptr, pooled.data_ptr(),
+                                   counts.data_ptr(), (int)B, (int)L, (int)D,
+                                   cur_stream()));
+  return {pooled, counts};
+}
+
+torch::Tensor masked_pool_bwd(torch::Tensor dpooled,
+                              c10::optional<torch::Tensor> mask,
+                              torch::Tensor counts, long L) {
+  check_bf16(dpooled, "dpooled");
+  long B = dpooled.size(0), D = dpooled.size(1);
+  const void* mptr = nullptr;
+  if (mask.has_value()) mptr = mask->data_ptr();
+  auto dx = torch::empty({B, L, D}, dpooled.options());
+  CHECK_HIP(masked_pool_bwd_launch(dpooled.data_ptr(), mptr,
+                                   counts.data_ptr(), dx.data_ptr(), (int)B,
+                                   (int)L, (int)D, cur_stream()));
+  return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("masked_pool_fwd", &masked_pool_fwd, "masked mean-pool fwd");
+  m.def("masked_pool_bwd", &masked_pool_bwd, "masked mean-pool bwd");
   m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
   m.def("flash_bwd_ds", &flash_bwd_ds, "fused flash bwd: P^T and dS^T tiles");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");

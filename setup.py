@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "csrc/adamw.hip",
         "csrc/repack.hip",
         "csrc/flash_attn.hip",
+        "csrc/pool.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

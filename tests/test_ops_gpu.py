@@ -294,3 +294,23 @@ def test_layernorm_bwd_with_residual_grad():
     dx0, _, _ = ops.hip_ops().layernorm_bwd(dyg, xg, gg, mean, rstd, None)
     expect = (dx0.float() + ds.float()).to(torch.bfloat16)
     assert torch.allclose(dx.float(), expect.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_masked_pool():
+    torch.manual_seed(16)
+    B, L, D = 4, 48, 128
+    x = _bf16(torch.randn(B, L, D))
+    mask = torch.zeros(B, L, dtype=torch.bool)
+    for i in range(B):
+        mask[i, :12 + 7 * i] = True
+    mg = mask.cuda().contiguous()
+    pooled, counts = ops.hip_ops().masked_pool_fwd(x, mg)
+    w = mask.float().unsqueeze(-1)
+    expect = (x.float().cpu() * w).sum(1) / w.sum(1)
+    assert torch.allclose(pooled.float().cpu(), expect, atol=2e-2, rtol=2e-2)
+    assert torch.equal(counts.cpu(), mask.sum(-1).float())
+    dp = _bf16(torch.randn(B, D))
+    dx = ops.hip_ops().masked_pool_bwd(dp, mg, counts, L)
+    dxe = (dp.float().cpu() / counts.cpu().unsqueeze(-1)).unsqueeze(1) \
+        .expand(-1, L, -1) * w
+    assert torch.allclose(dx.float().cpu(), dxe, atol=2e-2, rtol=2e-2)

@@ -197,12 +197,8 @@ class MLTC(nn.Module):
         else:
             x, _ = ops.fused_add_layernorm(x, delta, self.ln_f.weight,
                                            self.ln_f.bias, self.ln_f.eps)
-        # masked mean-pool over valid tokens
-        if attn_mask is not None:
-            w = attn_mask.to(x.dtype).unsqueeze(-1)
-            pooled = (x * w).sum(dim=1) / w.sum(dim=1).clamp(min=1)
-        else:
-            pooled = x.mean(dim=1)
+        # masked mean-pool over valid tokens (fused kernel: csrc/pool.hip)
+        pooled = ops.masked_mean_pool(x, attn_mask)
         return {name: head(pooled) for name, head in self.heads.items()}
 
     def set_pos_weights(self, weights: Dict[str, torch.Tensor]) -> None:

@@ -268,6 +268,48 @@ def flash_supported(head_dim: int, L: int) -> bool:
     return head_dim == 64 and L % 32 == 0
 
 
+class _MaskedPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, mask):
+        if x.is_cuda:
+            pooled, counts = hip_ops().masked_pool_fwd(x, mask)
+        else:
+            if mask is not None:
+                w = mask.to(x.dtype).unsqueeze(-1)
+                counts = mask.sum(-1).clamp(min=1).float()
+                pooled = ((x.float() * w.float()).sum(1) /
+                          counts.unsqueeze(-1)).to(x.dtype)
+            else:
+                counts = torch.full((x.shape[0],), x.shape[1],
+                                    dtype=torch.float32)
+                pooled = x.float().mean(1).to(x.dtype)
+        ctx.save_for_backward(counts)
+        ctx.mask = mask
+        ctx.L = x.shape[1]
+        return pooled
+
+    @staticmethod
+    def backward(ctx, dpooled):
+        (counts,) = ctx.saved_tensors
+        mask = ctx.mask
+        dpooled = dpooled.contiguous()
+        if dpooled.is_cuda:
+            dx = hip_ops().masked_pool_bwd(dpooled, mask, counts, ctx.L)
+        else:
+            g = (dpooled.float() / counts.unsqueeze(-1)).unsqueeze(1)
+            dx = g.expand(-1, ctx.L, -1)
+            if mask is not None:
+                dx = dx * mask.unsqueeze(-1).float()
+            dx = dx.to(dpooled.dtype)
+        return dx, None
+
+
+def masked_mean_pool(x, mask=None):
+    """pooled[b] = mean over valid rows of x[b]; x [B,L,D] bf16, mask [B,L]."""
+    return _MaskedPoolFn.apply(x.contiguous(),
+                               mask.contiguous() if mask is not None else None)
+
+
 def adamw_step(p, grad, m, v, master, *, lr, beta1=0.9, beta2=0.999, eps=1e-8,
                wd=0.01, step, grad_scale=1.0):
     """Fused AdamW over the flat parameter buffer (see train.py)."""
