@@ -59,3 +59,23 @@ def test_latest_checkpoint_picks_newest(tmp_path):
     open(os.path.join(ck, "ckpt_00000001.pt"), "w").close()
     open(os.path.join(ck, "ckpt_00000010.pt"), "w").close()
     assert Trainer.latest_checkpoint(ck).endswith("ckpt_00000010.pt")
+
+
+def test_gradient_accumulation_matches_big_batch():
+    torch.manual_seed(7)
+    cfg = CONFIGS["mltc-tiny"]
+    tokens, mask, labels = synthetic_batch(cfg, 8, 32, seed=21)
+
+    torch.manual_seed(50)
+    big = Trainer(TrainConfig(model="mltc-tiny", lr=1e-3, warmup_steps=0,
+                              dtype="f32"), device=torch.device("cpu"))
+    big.step(tokens, mask, labels)
+
+    torch.manual_seed(50)
+    acc = Trainer(TrainConfig(model="mltc-tiny", lr=1e-3, warmup_steps=0,
+                              dtype="f32"), device=torch.device("cpu"))
+    halves = [(tokens[:4], mask[:4], {k: v[:4] for k, v in labels.items()}),
+              (tokens[4:], mask[4:], {k: v[4:] for k, v in labels.items()})]
+    acc.step_accum(halves)
+    diff = (big.flat.flat - acc.flat.flat).abs().max()
+    assert float(diff) < 5e-5, float(diff)

@@ -46,6 +46,7 @@ class BucketedAllReduce:
             dist.get_world_size(process_group) > 1
         self.world_size = dist.get_world_size(process_group) if self.enabled else 1
         self.grad_flat = grad_flat
+        self.sync = True     # False during gradient-accumulation micro-steps
         self._buckets: List[_Bucket] = []
         self._param_bucket = {}
         self._hooks = []
@@ -93,6 +94,8 @@ class BucketedAllReduce:
             b.work = None
 
     def _on_grad_ready(self, param):
+        if not self.sync:
+            return
         b = self._param_bucket.get(id(param))
         if b is None:
             return

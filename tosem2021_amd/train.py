@@ -108,6 +108,34 @@ class Trainer:
         return c.lr * 0.5 * (1.0 + math.cos(math.pi * min(t, 1.0)))
 
     # ---- one optimization step ----------------------------------------------
+    def step_accum(self, micro_batches) -> float:
+        """One optimizer step over several micro-batches (gradient
+        accumulation): collectives fire only on the last micro-backward,
+        and the 1/n_micro mean folds into the AdamW grad scale."""
+        micro_batches = list(micro_batches)
+        n = len(micro_batches)
+        with trace("train_step_accum", step=self.step_num + 1, micro=n):
+            self.flat.zero_grad()
+            total_loss = 0.0
+            for i, (tokens, attn_mask, labels) in enumerate(micro_batches):
+                self.ddp.sync = (i == n - 1)
+                logits = self.model(tokens, attn_mask)
+                loss = self.model.loss(logits, labels)
+                loss.backward()
+                total_loss += float(loss.detach())
+            self.ddp.sync = True
+            self.ddp.finalize()
+            self.step_num += 1
+            lr = self._lr()
+            ops.adamw_step(
+                self.flat.flat, self.flat.grad_flat, self.flat.m, self.flat.v,
+                self.flat.master, lr=lr, beta1=self.cfg.beta1,
+                beta2=self.cfg.beta2, eps=self.cfg.eps,
+                wd=self.cfg.weight_decay, step=self.step_num,
+                grad_scale=self.ddp.grad_scale / n)
+        get_metrics().observe_step(self.step_num, loss=total_loss / n, lr=lr)
+        return total_loss / n
+
     def step(self, tokens: torch.Tensor, attn_mask: Optional[torch.Tensor],
              labels: Dict[str, torch.Tensor]) -> float:
         with trace("train_step", step=self.step_num + 1):
