@@ -314,3 +314,14 @@ def test_masked_pool():
     dxe = (dp.float().cpu() / counts.cpu().unsqueeze(-1)).unsqueeze(1) \
         .expand(-1, L, -1) * w
     assert torch.allclose(dx.float().cpu(), dxe, atol=2e-2, rtol=2e-2)
+
+
+def test_masked_pool_fully_masked_row():
+    x = _bf16(torch.randn(2, 16, 64))
+    mask = torch.zeros(2, 16, dtype=torch.bool)
+    mask[0, :5] = True  # row 1 fully masked
+    mg = mask.cuda().contiguous()
+    pooled, counts = ops.hip_ops().masked_pool_fwd(x, mg)
+    assert float(counts[1]) == 1.0  # clamped, no div-by-zero
+    assert torch.all(pooled[1].float() == 0)
+    assert not torch.isnan(pooled.float()).any()

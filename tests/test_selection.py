@@ -60,3 +60,17 @@ def test_reference_funnel_tables(reference_root):
     for org in ("ApolloAuto/apollo", "ray-project/ray", "microsoft/nni"):
         if org in set(v3["Repos"]):
             assert org in survivors, org
+
+
+def test_funnel_cli(reference_root, tmp_path):
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "tosem2021_amd.cli", "funnel", "--metrics",
+         os.path.join(reference_root, "selection", "Reposition",
+                      "Repos_metrics_v3.csv"),
+         "--out", str(tmp_path)],
+        capture_output=True, text=True, cwd=repo, timeout=300)
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "input: 311 candidates" in r.stdout
+    assert os.path.exists(tmp_path / "round_2.csv")

@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
 import sys
 
 
@@ -65,6 +66,28 @@ def cmd_report(args):
     print(summary_report(df))
 
 
+def cmd_funnel(args):
+    from tosem2021_amd.corpus.selection import (
+        DEFAULT_ROUNDS, FunnelCriteria, load_metrics, run_funnel)
+    df = load_metrics(args.metrics)
+    rounds = DEFAULT_ROUNDS
+    if args.min_stars or args.min_commits or args.min_contributors:
+        rounds = [FunnelCriteria(name="custom", min_stars=args.min_stars,
+                                 min_commits=args.min_commits,
+                                 min_contributors=args.min_contributors,
+                                 min_releases=args.min_releases)]
+    outs = run_funnel(df, rounds)
+    print(f"input: {len(df)} candidates")
+    for crit, out in zip(rounds, outs):
+        print(f"round '{crit.name}': {len(out)} survive")
+    if args.out:
+        os.makedirs(args.out, exist_ok=True)
+        for i, out in enumerate(outs):
+            p = os.path.join(args.out, f"round_{i + 1}.csv")
+            out.to_csv(p, index=False)
+            print(f"wrote {p}")
+
+
 def cmd_classify(args):
     from tosem2021_amd.classify.neural import apply_classifier
     out = apply_classifier(args.ckpt_dir, args.taxonomy, args.out,
@@ -114,6 +137,16 @@ def main(argv=None):
     p = sub.add_parser("report", help="print a taxonomy summary report")
     p.add_argument("--taxonomy", required=True)
     p.set_defaults(fn=cmd_report)
+
+    p = sub.add_parser("funnel", help="run the repo-selection funnel")
+    p.add_argument("--metrics", required=True,
+                   help="GitHub-metadata CSV (Repos_metrics_v3.csv schema)")
+    p.add_argument("--out", default=None)
+    p.add_argument("--min-stars", type=int, default=0)
+    p.add_argument("--min-commits", type=int, default=0)
+    p.add_argument("--min-contributors", type=int, default=0)
+    p.add_argument("--min-releases", type=int, default=0)
+    p.set_defaults(fn=cmd_funnel)
 
     p = sub.add_parser("classify", help="label a taxonomy with a trained model")
     p.add_argument("--taxonomy", required=True)
