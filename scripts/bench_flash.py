@@ -1,0 +1,47 @@
+"""Standalone flash-attention kernel microbench (for rocprofv3 PMC runs)."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import time
+
+import torch
+
+from tosem2021_amd import ops
+
+
+def main():
+    B, H, L, dh = 128, 16, 512, 64
+    iters = int(sys.argv[1]) if len(sys.argv) > 1 else 30
+    torch.manual_seed(0)
+    q = torch.randn(B, H, L, dh, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    mask = torch.zeros(B, L, device="cuda")
+    mask[:, 400:] = -1e9
+    mask = mask.contiguous()
+    scale = 0.125
+    ext = ops.hip_ops()
+    for _ in range(5):
+        o, lse = ext.flash_fwd(q, k, v, mask, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        o, lse = ext.flash_fwd(q, k, v, mask, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    flops = 4 * B * H * L * L * dh
+    print(f"flash_fwd {dt*1e6:.1f} us/call  {flops/dt/1e12:.1f} TF")
+    # bwd stage
+    do = torch.randn_like(q)
+    ddot = ext.fa_dot(do, o)
+    for _ in range(3):
+        p_t, ds_t = ext.flash_bwd_ds(q, k, v, do, mask, lse, ddot, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        p_t, ds_t = ext.flash_bwd_ds(q, k, v, do, mask, lse, ddot, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    print(f"flash_bwd_ds {dt*1e6:.1f} us/call  {2*flops/dt/1e12:.1f} TF-eq")
+
+
+if __name__ == "__main__":
+    main()
