@@ -118,7 +118,10 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
 
     // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
+    // (T5: raise wave priority through the MFMA cluster so co-resident
+    // waves' memory phases don't starve the matrix pipe)
     f32x16 s_acc = (f32x16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       // K fragment: lane holds K[kv=col][16c + 8*half + j] from swizzled LDS
@@ -128,6 +131,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                        kswz(row, byte_off));
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc, 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- online softmax (per lane: one q column, 16 kv rows) ----
     float sv[16];
@@ -190,6 +194,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
 
     // ---- PV: O[q, d] += P^T @ V — B-fragment from transposed V tile ----
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
@@ -203,6 +208,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                                            o_acc[t], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
@@ -330,6 +336,7 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
     // lane reads row q=col, feature chunk 16c + 8*half)
     f32x16 s_acc = (f32x16)(0.f);
     f32x16 dp_acc = (f32x16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int byte_off = (16 * c + 8 * half) * 2;
@@ -342,6 +349,7 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[c], dofrag, dp_acc,
                                                        0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // elementwise: P^T = exp(scale*S^T + bias - lse[q]); dS^T = scale*P*(dP-D)
     const int my_q = q0 + col;
