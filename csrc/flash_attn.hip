@@ -426,3 +426,45 @@ extern "C" hipError_t fa_dot_launch(const void* dout, const void* o,
                                           (float*)ddot, n_rows);
   return hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 semantics probe (development aid, exposed for tests):
+// stage `src` (bf16, 256 elements) into LDS, then each lane issues the
+// transpose-read at a caller-chosen address scheme and writes its 4 bf16
+// results to out[lane][0..3].
+//   scheme 0: addr = 0 for every lane
+//   scheme 1: addr = (lane & 15) * 2
+//   scheme 2: addr = ((lane & 15) + (lane >> 4) * 64) * 2
+//   scheme 3: addr = (lane >> 4) * 128  (16-lane-group base, uniform in group)
+extern "C" __global__ void __launch_bounds__(64)
+tr_probe_kernel(const short* __restrict__ src, short* __restrict__ out,
+                int scheme) {
+  __shared__ __attribute__((aligned(16))) short lds[256];
+  int lane = threadIdx.x;
+  for (int i = lane; i < 256; i += 64) lds[i] = src[i];
+  __syncthreads();
+  int addr;
+  switch (scheme) {
+    case 0: addr = 0; break;
+    case 1: addr = (lane & 15) * 2; break;
+    case 2: addr = ((lane & 15) + (lane >> 4) * 64) * 2; break;
+    default: addr = (lane >> 4) * 128; break;
+  }
+  typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
+  uint2_t r;
+  char* base = (char*)lds + addr;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(r) : "v"(base) : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  out[lane * 4 + 0] = (short)(r[0] & 0xffff);
+  out[lane * 4 + 1] = (short)(r[0] >> 16);
+  out[lane * 4 + 2] = (short)(r[1] & 0xffff);
+  out[lane * 4 + 3] = (short)(r[1] >> 16);
+}
+
+extern "C" hipError_t tr_probe_launch(const void* src, void* out, int scheme,
+                                      hipStream_t stream) {
+  tr_probe_kernel<<<1, 64, 0, stream>>>((const short*)src, (short*)out,
+                                        scheme);
+  return hipGetLastError();
+}

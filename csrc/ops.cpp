@@ -48,6 +48,7 @@ hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
                                   int, int, hipStream_t);
 hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
                                   void*, int, int, int, hipStream_t);
+hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
 }
 
 namespace {
@@ -405,7 +406,17 @@ torch::Tensor masked_pool_bwd(torch::Tensor dpooled,
   return dx;
 }
 
+torch::Tensor tr_probe(torch::Tensor src, long scheme) {
+  check_bf16(src, "src");
+  TORCH_CHECK(src.numel() == 256);
+  auto out = torch::zeros({64, 4}, src.options());
+  CHECK_HIP(tr_probe_launch(src.data_ptr(), out.data_ptr(), (int)scheme,
+                            cur_stream()));
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("tr_probe", &tr_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("masked_pool_fwd", &masked_pool_fwd, "masked mean-pool fwd");
   m.def("masked_pool_bwd", &masked_pool_bwd, "masked mean-pool bwd");
   m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
