@@ -453,7 +453,7 @@ tr_probe_kernel(const short* __restrict__ src, short* __restrict__ out,
   }
   typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
   uint2_t r;
-  char* base = (char*)lds + addr;
+  unsigned base = (unsigned)(unsigned long)((char*)lds + addr);
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                : "=v"(r) : "v"(base) : "memory");
   __builtin_amdgcn_sched_barrier(0);
