@@ -487,7 +487,10 @@ tr_probe_kernel(const short* __restrict__ src, short* __restrict__ out,
     case 0: addr = 0; break;
     case 1: addr = (lane & 15) * 2; break;
     case 2: addr = ((lane & 15) + (lane >> 4) * 64) * 2; break;
-    default: addr = (lane >> 4) * 128; break;
+    case 3: addr = (lane >> 4) * 128; break;
+    case 4: addr = (lane & 3) * 16; break;           // mates' ~3-bases differ
+    case 5: addr = ((lane & 3) * 32 + 4 * ((lane >> 2) & 7)) * 2; break;
+    default: addr = ((lane & 3) * 8 + 4) * 2; break; // aligned base + twist?
   }
   typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
   uint2_t r;
