@@ -79,3 +79,17 @@ def test_gradient_accumulation_matches_big_batch():
     acc.step_accum(halves)
     diff = (big.flat.flat - acc.flat.flat).abs().max()
     assert float(diff) < 5e-5, float(diff)
+
+
+def test_checkpoint_retention(tmp_path):
+    ck = str(tmp_path / "cks")
+    t = Trainer(TrainConfig(model="mltc-tiny", warmup_steps=0, ckpt_dir=ck,
+                            ckpt_keep=2), device=torch.device("cpu"))
+    cfg = CONFIGS["mltc-tiny"]
+    tokens, mask, labels = synthetic_batch(cfg, 2, 16, seed=3)
+    for _ in range(4):
+        t.step(tokens, mask, labels)
+        t.save()
+    kept = sorted(f for f in os.listdir(ck) if f.endswith(".pt"))
+    assert len(kept) == 2
+    assert kept[-1] == "ckpt_00000004.pt"

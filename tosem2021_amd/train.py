@@ -76,6 +76,7 @@ class TrainConfig:
     dtype: str = "bf16"  # "bf16" | "f32" (f32: CPU/gloo tests)
     ckpt_dir: Optional[str] = None
     ckpt_every: int = 0          # 0 = only on explicit save()
+    ckpt_keep: int = 3           # retain the newest k checkpoints (0 = all)
 
 
 class Trainer:
@@ -177,7 +178,22 @@ class Trainer:
         tmp = path + ".tmp"
         torch.save(self.state_dict(), tmp)
         os.replace(tmp, path)  # atomic publish (DeepSpeech-style durable ckpt)
+        self._prune(os.path.dirname(path))
         return path
+
+    def _prune(self, ckpt_dir: str):
+        """Keep the newest cfg.ckpt_keep checkpoints (tune checkpoint_manager
+        style retention)."""
+        k = self.cfg.ckpt_keep
+        if not k or not ckpt_dir:
+            return
+        cks = sorted(f for f in os.listdir(ckpt_dir)
+                     if f.startswith("ckpt_") and f.endswith(".pt"))
+        for f in cks[:-k]:
+            try:
+                os.remove(os.path.join(ckpt_dir, f))
+            except OSError:
+                pass
 
     def load(self, path: str):
         sd = torch.load(path, map_location=self.device, weights_only=True)
