@@ -21,8 +21,55 @@ from tosem2021_amd.train import TrainConfig, Trainer
 
 
 @torch.no_grad()
+def _head_probs(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
+                seq: int, batch: int = 64):
+    """Collect sigmoid probabilities and gold multi-hot labels per head."""
+    model = trainer.model
+    model.eval()
+    device = trainer.device
+    probs = {"strategy": [], "property": []}
+    gold = {"strategy": [], "property": []}
+    for toks, mask, labels in ds.batches(tok, batch, seq, device=device,
+                                         shuffle=False):
+        logits = model(toks, mask)
+        for h in probs:
+            probs[h].append(torch.sigmoid(logits[h].float()).cpu())
+            gold[h].append(labels[h].cpu())
+    model.train()
+    return ({h: torch.cat(v) for h, v in probs.items()},
+            {h: torch.cat(v) for h, v in gold.items()})
+
+
+def tune_thresholds(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
+                    seq: int) -> Dict[str, torch.Tensor]:
+    """Per-class decision thresholds maximizing F1 on `ds` (use the TRAIN
+    split here; evaluate() then reports on the held-out split)."""
+    probs, gold = _head_probs(trainer, ds, tok, seq)
+    grid = torch.arange(0.05, 0.95, 0.05)
+    out = {}
+    for h in probs:
+        p, g = probs[h], gold[h] > 0.5
+        ths = torch.full((p.shape[1],), 0.5)
+        for j in range(p.shape[1]):
+            best_f1, best_t = -1.0, 0.5
+            for t in grid:
+                pred = p[:, j] > t
+                tp = int((pred & g[:, j]).sum())
+                fp = int((pred & ~g[:, j]).sum())
+                fn = int((~pred & g[:, j]).sum())
+                f1 = 2 * tp / max(2 * tp + fp + fn, 1)
+                if f1 > best_f1:
+                    best_f1, best_t = f1, float(t)
+            ths[j] = best_t
+        out[h] = ths
+    return out
+
+
+@torch.no_grad()
 def evaluate(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
-             seq: int, batch: int = 64, threshold: float = 0.5) -> Dict[str, float]:
+             seq: int, batch: int = 64, threshold: float = 0.5,
+             thresholds: Optional[Dict[str, torch.Tensor]] = None
+             ) -> Dict[str, float]:
     model = trainer.model
     model.eval()
     device = trainer.device
@@ -33,8 +80,10 @@ def evaluate(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
     for toks, mask, labels in ds.batches(tok, batch, seq, device=device,
                                          shuffle=False):
         logits = model(toks, mask)
-        sp = (torch.sigmoid(logits["strategy"].float()) > threshold)
-        pp = (torch.sigmoid(logits["property"].float()) > threshold)
+        th_s = thresholds["strategy"].to(device) if thresholds else threshold
+        th_p = thresholds["property"].to(device) if thresholds else threshold
+        sp = (torch.sigmoid(logits["strategy"].float()) > th_s)
+        pp = (torch.sigmoid(logits["property"].float()) > th_p)
         mp = logits["method"].float().argmax(-1)
         gs = labels["strategy"] > 0.5
         gp = labels["property"] > 0.5
@@ -121,6 +170,10 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
     ev_lo = evaluate(trainer, val_ds, tok, seq, threshold=0.3)
     ev["strategy_micro_f1_t0.3"] = ev_lo["strategy_micro_f1"]
     ev["property_micro_f1_t0.3"] = ev_lo["property_micro_f1"]
+    ths = tune_thresholds(trainer, train_ds, tok, seq)
+    ev_tuned = evaluate(trainer, val_ds, tok, seq, thresholds=ths)
+    ev["strategy_micro_f1_tuned"] = round(ev_tuned["strategy_micro_f1"], 4)
+    ev["property_micro_f1_tuned"] = round(ev_tuned["property_micro_f1"], 4)
     if ckpt_dir:
         trainer.save()
     return {
