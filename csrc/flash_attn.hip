@@ -451,7 +451,16 @@ tr_probe_kernel(const short* __restrict__ src, short* __restrict__ out,
     case 3: addr = (lane >> 4) * 128; break;
     case 4: addr = (lane & 3) * 16; break;           // mates' ~3-bases differ
     case 5: addr = ((lane & 3) * 32 + 4 * ((lane >> 2) & 7)) * 2; break;
-    default: addr = ((lane & 3) * 8 + 4) * 2; break; // aligned base + twist?
+    case 6: addr = ((lane & 3) * 8 + 4) * 2; break;
+    // one-mate-at-a-time base perturbation (which mate feeds which output?)
+    case 7: addr = ((lane & 3) == 0) ? 16 : 0; break;
+    case 8: addr = ((lane & 3) == 1) ? 16 : 0; break;
+    case 9: addr = ((lane & 3) == 2) ? 16 : 0; break;
+    case 10: addr = ((lane & 3) == 3) ? 16 : 0; break;
+    // kernel-like: row stride 64 elems per mate + d-quad by quad index
+    case 11: addr = ((lane & 3) * 64 + 4 * ((lane >> 2) & 7)) * 2; break;
+    // distinct per mate AND per quad
+    default: addr = ((lane & 3) * 8 + ((lane >> 2) & 7) * 32) * 2; break;
   }
   typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
   uint2_t r;
