@@ -31,12 +31,13 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 #define FA_QWG (FA_WAVES * FA_QB)   // 128 q rows per workgroup
 
 // LDS layout (dynamic, 16-B aligned):
-//  K tile:  [32][64] bf16, row stride 128 B, XOR-swizzled      = 4096 B
-//  Vt tile: [64][32] bf16, row stride 80 B (pad 8)             = 5120 B
-//  alpha:   [4 waves][32] f32                                  =  512 B
+//  K tile: [32][64] bf16, row stride 128 B, XOR-swizzled       = 4096 B
+//  V tile: same layout — the PV B-fragment is gathered with
+//          ds_read_b64_tr_b16 (lane-grid-transpose semantics pinned by
+//          scripts/tr_probe.py; see the PV block below)         = 4096 B
+//  alpha:  [4 waves][32] f32                                   =  512 B
 #define K_LDS_BYTES (FA_KVB * 128)
-#define VT_ROW_BYTES 80
-#define VT_LDS_BYTES (FA_DH * VT_ROW_BYTES)
+#define VT_LDS_BYTES (FA_KVB * 128)
 
 __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
@@ -49,7 +50,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  int B, int H, int L, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = (short*)smem;                       // swizzled [32][64]
-  short* vt_lds = (short*)(smem + K_LDS_BYTES);      // [64][40] (80 B rows)
+  short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
   float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
 
   const int tid = threadIdx.x;
@@ -102,11 +103,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     __syncthreads();   // previous tile's LDS reads complete
     {
       *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int d = (sc8 >> 1) + j;
-        *(short*)((char*)vt_lds + d * VT_ROW_BYTES + srow * 2) = vv8[j];
-      }
+      *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
     }
     __syncthreads();   // tile t visible to all waves
     if (kt + 1 < n_kv) {
@@ -193,22 +190,67 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       pf[c] = __builtin_bit_cast(short8_t, u);
     }
 
-    // ---- PV: O[q, d] += P^T @ V — B-fragment from transposed V tile ----
-    __builtin_amdgcn_s_setprio(1);
+    // ---- PV: O[q, d] += P^T @ V ----
+    // B-fragment (lane: V[kv = 16c + 8*half + j][d = 32t + (l&31)], j=0..7)
+    // gathered by ds_read_b64_tr_b16 from the K-style swizzled row-major V
+    // tile.  Empirical semantics (scripts/tr_probe.py, 13 address schemes):
+    // within each 16-lane group, out[l][j] = lds[A(src) + (l & 3)] with
+    // src = (l & ~15) + 4*j + ((l >> 2) & 3)   — a 4x4 transpose of the
+    // group's (quad, mate) lane grid.  So lane l supplies the base of
+    //   V[kv0 + ((l>>2)&3)][32t + 16*((l>>4)&1) + 4*(l&3)]
+    // and receives column (l&3) across j = quad rows kv0..kv0+3.
+    {
+      typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
+      const unsigned vbase = (unsigned)(unsigned long)(char*)v_lds;
+      const int kv_mate = (lane >> 2) & 3;
+      const int d_lane = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+      unsigned a[8];
 #pragma unroll
-    for (int t = 0; t < 2; ++t) {
+      for (int c = 0; c < 2; ++c)
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        // lane holds V[kv = 16c + 8*half + j][d = 32t + col]
-        int d = 32 * t + col;
-        int kvb = 16 * c + 8 * half;
-        short8_t vf = *(const short8_t*)((char*)vt_lds + d * VT_ROW_BYTES +
-                                         kvb * 2);
-        o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
-                                                           o_acc[t], 0, 0, 0);
+        for (int rr = 0; rr < 2; ++rr)
+#pragma unroll
+          for (int t = 0; t < 2; ++t) {
+            int kv = 16 * c + 8 * half + 4 * rr + kv_mate;
+            int dcol = (32 * t + d_lane) * 2;
+            a[c * 4 + rr * 2 + t] =
+                vbase + kv * 128 + (dcol ^ ((kv & 7) << 4));
+          }
+      uint2_t r[8];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8\n\t"
+          "ds_read_b64_tr_b16 %1, %9\n\t"
+          "ds_read_b64_tr_b16 %2, %10\n\t"
+          "ds_read_b64_tr_b16 %3, %11\n\t"
+          "ds_read_b64_tr_b16 %4, %12\n\t"
+          "ds_read_b64_tr_b16 %5, %13\n\t"
+          "ds_read_b64_tr_b16 %6, %14\n\t"
+          "ds_read_b64_tr_b16 %7, %15\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(r[0]), "=&v"(r[1]), "=&v"(r[2]), "=&v"(r[3]),
+            "=&v"(r[4]), "=&v"(r[5]), "=&v"(r[6]), "=&v"(r[7])
+          : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3]), "v"(a[4]), "v"(a[5]),
+            "v"(a[6]), "v"(a[7])
+          : "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+          uint4_t w;
+          w[0] = r[c * 4 + 0 * 2 + t][0];
+          w[1] = r[c * 4 + 0 * 2 + t][1];
+          w[2] = r[c * 4 + 1 * 2 + t][0];
+          w[3] = r[c * 4 + 1 * 2 + t][1];
+          short8_t vf = __builtin_bit_cast(short8_t, w);
+          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
+                                                             o_acc[t], 0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
