@@ -5,9 +5,17 @@
 The resulting _hip_ops*.so lands in tosem2021_amd/ (in-tree, so it travels
 with the repo snapshot to GPU boxes; it is git-ignored).
 """
+import glob
 import os
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+# torch's hipify writes generated csrc/*_hip.hip next to the sources and can
+# SKIP regeneration even when the source is newer — delete them up front so
+# every build compiles the current kernels.
+for _f in glob.glob(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                 "csrc", "*_hip.hip")):
+    os.remove(_f)
 
 from setuptools import setup
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
