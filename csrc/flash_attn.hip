@@ -49,10 +49,9 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  short* __restrict__ o, float* __restrict__ lse,
                  int B, int H, int L, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered tiles: one __syncthreads per tile (write the NEXT tile
-  // into the other buffer after compute, barrier once).  Buffer p occupies
-  // smem + p*2*K_LDS_BYTES (K tile) and + K_LDS_BYTES (V tile).
-  float* alpha_lds = (float*)(smem + 4 * K_LDS_BYTES);
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
+  float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -99,23 +98,21 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                     (sc8 >> 1));
   short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
                                     (sc8 >> 1));
-  // prologue: tile 0 into buffer 0
-  {
-    *(short8_t*)(smem + srow * 128 + kswz(srow, sc8)) = kv8;
-    *(short8_t*)(smem + K_LDS_BYTES + srow * 128 + kswz(srow, sc8)) = vv8;
-  }
-  if (n_kv > 1) {
-    kv8 = *(const short8_t*)(k + bh_off + (long)(FA_KVB + srow) * FA_DH +
-                             (sc8 >> 1));
-    vv8 = *(const short8_t*)(v + bh_off + (long)(FA_KVB + srow) * FA_DH +
-                             (sc8 >> 1));
-  }
-  __syncthreads();
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
-    char* buf = smem + (kt & 1) * 2 * K_LDS_BYTES;
-    short* k_lds = (short*)buf;
-    short* v_lds = (short*)(buf + K_LDS_BYTES);
+    __syncthreads();   // previous tile's LDS reads complete
+    {
+      *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
+      *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
+    }
+    __syncthreads();   // tile t visible to all waves
+    if (kt + 1 < n_kv) {
+      // issue tile t+1 loads now; they stay in flight under the MFMAs
+      kv8 = *(const short8_t*)(k + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+    }
 
     // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
     // (T5: raise wave priority through the MFMA cluster so co-resident
@@ -281,7 +278,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        hipStream_t stream) {
   int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
   dim3 grid(B * H * n_qblocks);
-  size_t shm = 4 * K_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale);
