@@ -43,7 +43,7 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
