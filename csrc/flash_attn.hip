@@ -304,7 +304,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
 // Each wave owns one 32-row KV block (K/V fragments live in registers for the
 // whole kernel); the workgroup's 4 waves share the staged Q/dO tiles.
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
 flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                     const short* __restrict__ v, const short* __restrict__ dout,
                     const float* __restrict__ mask,
