@@ -43,11 +43,15 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
                  int B, int H, int L, float scale) {
+  // Each wave processes TWO independent 32-row q-blocks against the shared
+  // K/V tile: the second block's MFMAs overlap the first block's serial
+  // softmax chain (ILP within the wave).  2 waves/SIMD by registers; the
+  // workgroup covers 256 q rows.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = (short*)smem;                       // swizzled [32][64]
   short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
@@ -56,149 +60,147 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
-  const int col = lane & 31;       // q-row (QK^T) / d-col (PV) of this lane
+  const int col = lane & 31;
   const int half = lane >> 5;
 
-  // grid: x = (b*H + h) * n_qblocks + qb; remapped so one (b,h)'s q-blocks
-  // share an XCD and re-read K/V from that XCD's L2 (guide T1)
-  const int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
+  const int rows_per_wg = 2 * FA_QWG;                 // 256
+  const int n_qblocks = (L + rows_per_wg - 1) / rows_per_wg;
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
   int bh = bid / n_qblocks;
   int qb = bid % n_qblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
-  const int q_base_wg = qb * FA_QWG;
-  const int q_base = q_base_wg + wid * FA_QB;        // this wave's 32 q rows
-  const int my_q = q_base + col;                      // this lane's q row
-  const bool q_valid = my_q < L;
+  // wave's two q-blocks: rows [qA, qA+32) and [qB, qB+32)
+  const int q_baseA = qb * rows_per_wg + wid * FA_QB;
+  const int q_baseB = q_baseA + FA_QWG;
+  const int my_qA = q_baseA + col;
+  const int my_qB = q_baseB + col;
+  const bool validA = my_qA < L;
+  const bool validB = my_qB < L;
   const float* mrow = mask ? mask + (long)b * L : nullptr;
 
-  // ---- load Q fragments (once): A/B-operand layout, 4 k-chunks of 16 ----
-  // lane holds Q[my_q][16c + 8*half + j], j=0..8
-  short8_t qf[4];
+  short8_t qfA[4], qfB[4];
   {
-    const short* qr = q + bh_off + (long)(q_valid ? my_q : L - 1) * FA_DH;
+    const short* qrA = q + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
+    const short* qrB = q + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
 #pragma unroll
-    for (int c = 0; c < 4; ++c)
-      qf[c] = *(const short8_t*)(qr + c * 16 + half * 8);
+    for (int c = 0; c < 4; ++c) {
+      qfA[c] = *(const short8_t*)(qrA + c * 16 + half * 8);
+      qfB[c] = *(const short8_t*)(qrB + c * 16 + half * 8);
+    }
   }
 
-  f32x16 o_acc[2];                 // O[d-tile t][16 q-rows], d = 32t + col
+  f32x16 oA[2], oB[2];
 #pragma unroll
-  for (int t = 0; t < 2; ++t) o_acc[t] = (f32x16)(0.f);
-  float m_run = -3.0e38f;
-  float l_run = 0.f;
+  for (int t = 0; t < 2; ++t) { oA[t] = (f32x16)(0.f); oB[t] = (f32x16)(0.f); }
+  float mA = -3.0e38f, lA = 0.f, mB = -3.0e38f, lB = 0.f;
 
-  // T14 async-stage split: tile t+1's global loads are issued before tile
-  // t's compute and land in registers while the MFMAs run; the LDS write of
-  // tile t happens between the barrier pair (write-late, guide G15).
-  const int n_kv = L / FA_KVB;     // host asserts L % 32 == 0
-  const int srow = tid >> 3, sc8 = (tid & 7) * 16;  // this thread's 16-B piece
+  const int n_kv = L / FA_KVB;
+  const int srow = tid >> 3, sc8 = (tid & 7) * 16;
   short8_t kv8 = *(const short8_t*)(k + bh_off + (long)srow * FA_DH +
                                     (sc8 >> 1));
   short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
                                     (sc8 >> 1));
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
-    __syncthreads();   // previous tile's LDS reads complete
+    __syncthreads();
     {
       *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
       *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
     }
-    __syncthreads();   // tile t visible to all waves
+    __syncthreads();
     if (kt + 1 < n_kv) {
-      // issue tile t+1 loads now; they stay in flight under the MFMAs
       kv8 = *(const short8_t*)(k + bh_off +
                                (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
       vv8 = *(const short8_t*)(v + bh_off +
                                (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
     }
 
-    // ---- S^T tile: D[kv, q] = K @ Q^T, accumulate over 4 k-chunks ----
-    // (T5: raise wave priority through the MFMA cluster so co-resident
-    // waves' memory phases don't starve the matrix pipe)
-    f32x16 s_acc = (f32x16)(0.f);
+    // ---- QK^T for BOTH q-blocks (8 back-to-back MFMAs) ----
+    f32x16 sA = (f32x16)(0.f), sB = (f32x16)(0.f);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
-      // K fragment: lane holds K[kv=col][16c + 8*half + j] from swizzled LDS
-      int row = col;
       int byte_off = (16 * c + 8 * half) * 2;
-      short8_t kf = *(const short8_t*)((char*)k_lds + row * 128 +
-                                       kswz(row, byte_off));
-      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], s_acc, 0, 0, 0);
+      short8_t kf = *(const short8_t*)((char*)k_lds + col * 128 +
+                                       kswz(col, byte_off));
+      sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfA[c], sA, 0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfB[c], sB, 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- online softmax (per lane: one q column, 16 kv rows) ----
-    float sv[16];
-    float tmax = -3.0e38f;
+    // ---- online softmax, both blocks (independent chains -> ILP) ----
+    float svA[16], svB[16];
+    float tmaxA = -3.0e38f, tmaxB = -3.0e38f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float x = s_acc[r] * scale;
-      if (mrow) x += mrow[kv0 + kv_local];
-      sv[r] = x;
-      tmax = fmaxf(tmax, x);
+      float mb = mrow ? mrow[kv0 + kv_local] : 0.f;
+      float xA = sA[r] * scale + mb;
+      float xB = sB[r] * scale + mb;
+      svA[r] = xA; svB[r] = xB;
+      tmaxA = fmaxf(tmaxA, xA);
+      tmaxB = fmaxf(tmaxB, xB);
     }
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
-    float m_new = fmaxf(m_run, tmax);
-    float a = __expf(m_run - m_new);
-    float rsum = 0.f;
+    tmaxA = fmaxf(tmaxA, __shfl_xor(tmaxA, 32, WAVE));
+    tmaxB = fmaxf(tmaxB, __shfl_xor(tmaxB, 32, WAVE));
+    float mnA = fmaxf(mA, tmaxA), mnB = fmaxf(mB, tmaxB);
+    float aA = __expf(mA - mnA), aB = __expf(mB - mnB);
+    float rsA = 0.f, rsB = 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      sv[r] = __expf(sv[r] - m_new);
-      rsum += sv[r];
+      svA[r] = __expf(svA[r] - mnA);
+      svB[r] = __expf(svB[r] - mnB);
+      rsA += svA[r];
+      rsB += svB[r];
     }
-    rsum += __shfl_xor(rsum, 32, WAVE);
-    l_run = l_run * a + rsum;
-    m_run = m_new;
+    rsA += __shfl_xor(rsA, 32, WAVE);
+    rsB += __shfl_xor(rsB, 32, WAVE);
+    lA = lA * aA + rsA; mA = mnA;
+    lB = lB * aB + rsB; mB = mnB;
 
-    // broadcast alpha(q) to the O accumulator rows via per-wave LDS
-    // (both halves write the same value; wave-internal ds ordering suffices)
-    alpha_lds[wid * 32 + col] = a;
+    // alpha broadcast for both blocks (two slots per wave)
+    alpha_lds[wid * 64 + col] = aA;
+    alpha_lds[wid * 64 + 32 + col] = aB;
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        o_acc[t][r] *= alpha_lds[wid * 32 + qrow];
+        oA[t][r] *= alpha_lds[wid * 64 + qrow];
+        oB[t][r] *= alpha_lds[wid * 64 + 32 + qrow];
       }
     }
 
-    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap, guide T12) ----
-    // chunk 0: kv 8*half..+8 from regs 0..7; chunk 1: kv 16+8*half from 8..15
-    short8_t pf[2];
+    // ---- P -> bf16 fragments, both blocks ----
+    short8_t pfA[2], pfB[2];
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
       typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
-      uint4_t u;
+      uint4_t uA, uB;
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
-        int r0 = c * 8 + 2 * i;         // own pair (kv low quad)
-        int r1 = c * 8 + 4 + 2 * i;     // pair the partner half needs (+8)
-        unsigned lo, hi;
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
-            : "v"(sv[r0]), "v"(sv[r0 + 1]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
-            : "v"(sv[r1]), "v"(sv[r1 + 1]));
-        auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
-        // consecutive-kv order: {01, 23, 45, 67} (own pairs i=0,1 first)
-        u[i] = sw[0];
-        u[i + 2] = sw[1];
+        int r0 = c * 8 + 2 * i;
+        int r1 = c * 8 + 4 + 2 * i;
+        unsigned loA, hiA, loB, hiB;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(loA)
+            : "v"(svA[r0]), "v"(svA[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hiA)
+            : "v"(svA[r1]), "v"(svA[r1 + 1]));
+        auto swA = __builtin_amdgcn_permlane32_swap(loA, hiA, false, false);
+        uA[i] = swA[0]; uA[i + 2] = swA[1];
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(loB)
+            : "v"(svB[r0]), "v"(svB[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hiB)
+            : "v"(svB[r1]), "v"(svB[r1 + 1]));
+        auto swB = __builtin_amdgcn_permlane32_swap(loB, hiB, false, false);
+        uB[i] = swB[0]; uB[i + 2] = swB[1];
       }
-      pf[c] = __builtin_bit_cast(short8_t, u);
+      pfA[c] = __builtin_bit_cast(short8_t, uA);
+      pfB[c] = __builtin_bit_cast(short8_t, uB);
     }
 
-    // ---- PV: O[q, d] += P^T @ V ----
-    // B-fragment (lane: V[kv = 16c + 8*half + j][d = 32t + (l&31)], j=0..7)
-    // gathered by ds_read_b64_tr_b16 from the K-style swizzled row-major V
-    // tile.  Empirical semantics (scripts/tr_probe.py, 13 address schemes):
-    // within each 16-lane group, out[l][j] = lds[A(src) + (l & 3)] with
-    // src = (l & ~15) + 4*j + ((l >> 2) & 3)   — a 4x4 transpose of the
-    // group's (quad, mate) lane grid.  So lane l supplies the base of
-    //   V[kv0 + ((l>>2)&3)][32t + 16*((l>>4)&1) + 4*(l&3)]
-    // and receives column (l&3) across j = quad rows kv0..kv0+3.
+    // ---- PV for both blocks from tr_b16-gathered V fragments ----
     {
       typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
       const unsigned vbase = (unsigned)(unsigned long)(char*)v_lds;
@@ -245,28 +247,35 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
           w[2] = r[c * 4 + 1 * 2 + t][0];
           w[3] = r[c * 4 + 1 * 2 + t][1];
           short8_t vf = __builtin_bit_cast(short8_t, w);
-          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], vf,
-                                                             o_acc[t], 0, 0, 0);
+          oA[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pfA[c], vf,
+                                                          oA[t], 0, 0, 0);
+          oB[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pfB[c], vf,
+                                                          oB[t], 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
     }
   }
 
-  // ---- epilogue: O /= l (per REG row, via the alpha broadcast slot),
-  //      store O rows + logsumexp ----
-  if (q_valid && lse != nullptr && half == 0)
-    lse[(long)bh * L + my_q] = m_run + __logf(l_run);
-  alpha_lds[wid * 32 + col] = 1.0f / l_run;
+  // ---- epilogue: both blocks ----
+  if (lse != nullptr && half == 0) {
+    if (validA) lse[(long)bh * L + my_qA] = mA + __logf(lA);
+    if (validB) lse[(long)bh * L + my_qB] = mB + __logf(lB);
+  }
+  alpha_lds[wid * 64 + col] = 1.0f / lA;
+  alpha_lds[wid * 64 + 32 + col] = 1.0f / lB;
 #pragma unroll
   for (int t = 0; t < 2; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
-      int qrow = q_base + rloc;
-      if (qrow >= L) continue;
-      o[bh_off + (long)qrow * FA_DH + 32 * t + col] =
-          f32_to_bf16(o_acc[t][r] * alpha_lds[wid * 32 + rloc]);
+      int qA = q_baseA + rloc, qB = q_baseB + rloc;
+      if (qA < L)
+        o[bh_off + (long)qA * FA_DH + 32 * t + col] =
+            f32_to_bf16(oA[t][r] * alpha_lds[wid * 64 + rloc]);
+      if (qB < L)
+        o[bh_off + (long)qB * FA_DH + 32 * t + col] =
+            f32_to_bf16(oB[t][r] * alpha_lds[wid * 64 + 32 + rloc]);
     }
   }
 }
@@ -276,14 +285,15 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        void* o, void* lse, int B, int H,
                                        int L, float scale,
                                        hipStream_t stream) {
-  int n_qblocks = (L + FA_QWG - 1) / FA_QWG;
+  int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_qblocks);
-  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 32 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale);
   return hipGetLastError();
 }
+
 
 
 // ---------------------------------------------------------------------------
