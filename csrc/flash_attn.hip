@@ -314,7 +314,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
 // Each wave owns one 32-row KV block (K/V fragments live in registers for the
 // whole kernel); the workgroup's 4 waves share the staged Q/dO tiles.
 
-extern "C" __global__ void __launch_bounds__(FA_BLOCK, 3)
+extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
 flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                     const short* __restrict__ v, const short* __restrict__ dout,
                     const float* __restrict__ mask,
@@ -322,9 +322,14 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                     const float* __restrict__ ddot,
                     short* __restrict__ p_t, short* __restrict__ ds_t,
                     int B, int H, int L, float scale) {
+  // Each wave owns TWO 32-row KV blocks (K/V fragments resident for both):
+  // the second block's MFMA/elementwise chains overlap the first block's
+  // serial work.  2 waves/SIMD by registers; the workgroup covers 256 kv
+  // rows.  Mask bias is staged in LDS (registers are the scarce resource).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = (short*)smem;                        // swizzled [32][64]
   short* do_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
+  float* mb_lds = (float*)(smem + 2 * K_LDS_BYTES);   // [256] kv mask bias
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -332,35 +337,39 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int col = lane & 31;        // q column of this lane
   const int half = lane >> 5;
 
-  const int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;   // 128 kv rows per WG
+  const int rows_per_wg = 2 * FA_QWG;                 // 256 kv rows per WG
+  const int n_kvblocks = (L + rows_per_wg - 1) / rows_per_wg;
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_kvblocks);
   int bh = bid / n_kvblocks;
   int kb = bid % n_kvblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
   const long bh_sq = (long)bh * L * L;
-  const int kv_base = kb * FA_QWG + wid * FA_KVB;     // this wave's 32 kv rows
-  const int my_kv = kv_base + col;
-  const bool kv_valid = my_kv < L;
+  const int kv_wg = kb * rows_per_wg;
+  const int kv_baseA = kv_wg + wid * FA_KVB;          // wave's first kv block
+  const int kv_baseB = kv_baseA + FA_QWG;             // second, 128 later
   const float* mrow = mask ? mask + (long)b * L : nullptr;
 
-  // K and V fragments for this wave's kv block (resident all kernel)
-  short8_t kf[4], vf[4];
+  // stage the WG's 256 mask-bias values once
+  if (tid < rows_per_wg)
+    mb_lds[tid] = (mrow && kv_wg + tid < L) ? mrow[kv_wg + tid] : 0.f;
+
+  // K and V fragments for both kv blocks (resident all kernel)
+  short8_t kfA[4], vfA[4], kfB[4], vfB[4];
   {
-    const short* kr = k + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
-    const short* vr = v + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
+    const int mkA = min(kv_baseA + col, L - 1);
+    const int mkB = min(kv_baseB + col, L - 1);
+    const short* krA = k + bh_off + (long)mkA * FA_DH;
+    const short* vrA = v + bh_off + (long)mkA * FA_DH;
+    const short* krB = k + bh_off + (long)mkB * FA_DH;
+    const short* vrB = v + bh_off + (long)mkB * FA_DH;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
-      kf[c] = *(const short8_t*)(kr + c * 16 + half * 8);
-      vf[c] = *(const short8_t*)(vr + c * 16 + half * 8);
+      kfA[c] = *(const short8_t*)(krA + c * 16 + half * 8);
+      vfA[c] = *(const short8_t*)(vrA + c * 16 + half * 8);
+      kfB[c] = *(const short8_t*)(krB + c * 16 + half * 8);
+      vfB[c] = *(const short8_t*)(vrB + c * 16 + half * 8);
     }
-  }
-  // per-reg mask bias of this wave's kv rows (kv = kv_base + crow(r, half))
-  float mbias[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
-    mbias[r] = mrow ? mrow[min(kv_base + kvl, L - 1)] : 0.f;
   }
 
   const int n_q = L / FA_KVB;
@@ -384,10 +393,9 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                (sc8 >> 1));
     }
 
-    // S^T = K @ Q^T and dP^T = V @ dO^T (B-fragments from the staged tiles:
-    // lane reads row q=col, feature chunk 16c + 8*half)
-    f32x16 s_acc = (f32x16)(0.f);
-    f32x16 dp_acc = (f32x16)(0.f);
+    // S^T = K @ Q^T and dP^T = V @ dO^T for BOTH kv blocks (16 MFMAs)
+    f32x16 sA = (f32x16)(0.f), dpA = (f32x16)(0.f);
+    f32x16 sB = (f32x16)(0.f), dpB = (f32x16)(0.f);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
@@ -396,41 +404,49 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                           kswz(col, byte_off));
       short8_t dofrag = *(const short8_t*)((char*)do_lds + col * 128 +
                                            kswz(col, byte_off));
-      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[c], qfrag, s_acc,
-                                                      0, 0, 0);
-      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[c], dofrag, dp_acc,
-                                                       0, 0, 0);
+      sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[c], qfrag, sA, 0, 0, 0);
+      dpA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[c], dofrag, dpA,
+                                                    0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfB[c], qfrag, sB, 0, 0, 0);
+      dpB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfB[c], dofrag, dpB,
+                                                    0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // elementwise: P^T = exp(scale*S^T + bias - lse[q]); dS^T = scale*P*(dP-D)
+    // elementwise + store, block A then block B (independent chains)
     const int my_q = q0 + col;
     const float l_q = lse[(long)bh * L + my_q];
     const float d_q = ddot[(long)bh * L + my_q];
-    short8_t p_pack[2], ds_pack[2];
 #pragma unroll
-    for (int rr = 0; rr < 16; rr += 2) {
-      float p0 = __expf(s_acc[rr] * scale + mbias[rr] - l_q);
-      float p1 = __expf(s_acc[rr + 1] * scale + mbias[rr + 1] - l_q);
-      float g0 = scale * p0 * (dp_acc[rr] - d_q);
-      float g1 = scale * p1 * (dp_acc[rr + 1] - d_q);
-      int slot = rr >> 1;  // 8 packed pairs
-      p_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(p0);
-      p_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(p1);
-      ds_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(g0);
-      ds_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(g1);
-    }
-    // store: rows kv = kv_base + crow(2*slot..), cols q0 + col.
-    // Row pairs (2*slot, 2*slot+1) are CONSECUTIVE kv rows, so store the two
-    // bf16 of a pair as separate row stores (2 B per lane, coalesced per row).
+    for (int blk = 0; blk < 2; ++blk) {
+      const f32x16& s_acc = blk ? sB : sA;
+      const f32x16& dp_acc = blk ? dpB : dpA;
+      const int kv_base = blk ? kv_baseB : kv_baseA;
+      const int mb0 = kv_base - kv_wg;
+      short8_t p_pack[2], ds_pack[2];
 #pragma unroll
-    for (int rr = 0; rr < 16; ++rr) {
-      int kvl = (rr & 3) + 8 * (rr >> 2) + 4 * half;
-      if (kv_base + kvl >= L) continue;  // tail kv-block: no OOB rows
-      long off = bh_sq + (long)(kv_base + kvl) * L + q0 + col;
-      int slot = rr >> 1;
-      p_t[off] = p_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
-      ds_t[off] = ds_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+      for (int rr = 0; rr < 16; rr += 2) {
+        int kl0 = (rr & 3) + 8 * (rr >> 2) + 4 * half;
+        float p0 = __expf(s_acc[rr] * scale + mb_lds[mb0 + kl0] - l_q);
+        float p1 = __expf(s_acc[rr + 1] * scale + mb_lds[mb0 + kl0 + 1] - l_q);
+        float g0 = scale * p0 * (dp_acc[rr] - d_q);
+        float g1 = scale * p1 * (dp_acc[rr + 1] - d_q);
+        int slot = rr >> 1;  // 8 packed pairs
+        p_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(p0);
+        p_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(p1);
+        ds_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(g0);
+        ds_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(g1);
+      }
+      // store: rows kv = kv_base + crow(2*slot..), cols q0 + col (2 B/lane)
+#pragma unroll
+      for (int rr = 0; rr < 16; ++rr) {
+        int kvl = (rr & 3) + 8 * (rr >> 2) + 4 * half;
+        if (kv_base + kvl >= L) continue;  // tail kv-block: no OOB rows
+        long off = bh_sq + (long)(kv_base + kvl) * L + q0 + col;
+        int slot = rr >> 1;
+        p_t[off] = p_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+        ds_t[off] = ds_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+      }
     }
   }
 }
@@ -439,15 +455,16 @@ extern "C" hipError_t flash_bwd_ds_launch(
     const void* q, const void* k, const void* v, const void* dout,
     const void* mask, const void* lse, const void* ddot, void* p_t,
     void* ds_t, int B, int H, int L, float scale, hipStream_t stream) {
-  int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;
+  int n_kvblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_kvblocks);
-  size_t shm = 2 * K_LDS_BYTES;
+  size_t shm = 2 * K_LDS_BYTES + 2 * FA_QWG * sizeof(float);
   flash_bwd_ds_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)mask, (const float*)lse, (const float*)ddot, (short*)p_t,
       (short*)ds_t, B, H, L, scale);
   return hipGetLastError();
 }
+
 
 // D = rowsum(dO * O) per (b, h, q) row — one wave per 4 rows (dh = 64).
 extern "C" __global__ void __launch_bounds__(256)
