@@ -70,6 +70,25 @@ def test_train_classifier_tiny_cpu(tmp_path):
     assert any(f.startswith("ckpt_") for f in os.listdir(tmp_path / "ck"))
 
 
+def test_train_classifier_with_pretrain_cpu(tmp_path):
+    from tosem2021_amd.classify.neural import train_classifier
+    tax = str(tmp_path / "gold.csv")
+    pre = str(tmp_path / "mined.csv")
+    _write_tiny_taxonomy(tax)
+    _write_tiny_taxonomy(pre, n=64)
+    res = train_classifier(tax, model="mltc-tiny", steps=4, batch=8, seq=32,
+                           lr=1e-3, device="cpu",
+                           pretrain_path=pre, pretrain_steps=3)
+    assert res["steps"] == 4
+    assert res["pretrain_steps"] == 3
+    assert res["pretrain_time_s"] > 0
+    assert res["final_loss"] == res["final_loss"]  # not NaN
+    # warm start must actually change the outcome vs cold start at equal seed
+    cold = train_classifier(tax, model="mltc-tiny", steps=4, batch=8, seq=32,
+                            lr=1e-3, device="cpu")
+    assert res["final_loss"] != cold["final_loss"]
+
+
 def test_linear_baseline_tiny(tmp_path):
     from tosem2021_amd.classify.baseline import train_linear_baseline
     tax = str(tmp_path / "tiny_tax.csv")

@@ -108,33 +108,73 @@ def evaluate(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
     }
 
 
+def _load_dataset(taxonomy_path: str, cfg: MLTCConfig) -> TaxonomyDataset:
+    if taxonomy_path.endswith(".pt"):
+        ds = TaxonomyDataset.from_tensor_file(taxonomy_path)
+    else:
+        df = load_taxonomy(taxonomy_path)
+        ds = TaxonomyDataset.from_taxonomy(df)
+    if ds._tokens is not None and int(ds._tokens.max()) >= cfg.vocab_size:
+        # prepared file was tokenized for a bigger vocab (e.g. mltc-base);
+        # fold ids into this config's hash space deterministically
+        from tosem2021_amd.models.tokenizer import N_RESERVED
+        t = ds._tokens
+        big = t >= N_RESERVED
+        ds._tokens = torch.where(
+            big, (t - N_RESERVED) % (cfg.vocab_size - N_RESERVED) + N_RESERVED,
+            t)
+    return ds
+
+
+def _run_steps(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
+               batch: int, seq: int, steps: int, seed: int, device,
+               eval_every: int = 0, val_ds: Optional[TaxonomyDataset] = None,
+               tag: str = "") -> list:
+    losses = []
+    epoch = 0
+    it = iter(())
+    while trainer.step_num < steps:
+        try:
+            toks, mask, labels = next(it)
+        except StopIteration:
+            it = ds.batches(tok, batch, seq, device=device, shuffle=True,
+                            seed=seed + epoch, drop_last=True)
+            epoch += 1
+            continue
+        losses.append(trainer.step(toks, mask, labels))
+        if eval_every and val_ds is not None and \
+                trainer.step_num % eval_every == 0:
+            ev = evaluate(trainer, val_ds, tok, seq)
+            print(f"{tag}step {trainer.step_num} loss {losses[-1]:.4f} "
+                  f"strategyF1 {ev['strategy_micro_f1']:.3f} "
+                  f"propF1 {ev['property_micro_f1']:.3f} "
+                  f"methodAcc {ev['method_accuracy']:.3f}")
+    return losses, epoch
+
+
 def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                      steps: int = 500, batch: int = 32, seq: int = 256,
                      lr: float = 3e-4, ckpt_dir: Optional[str] = None,
                      resume: bool = False, device: Optional[str] = None,
                      eval_every: int = 0, seed: int = 0,
-                     dropout: float = 0.0) -> dict:
+                     dropout: float = 0.0,
+                     pretrain_path: Optional[str] = None,
+                     pretrain_steps: int = 0) -> dict:
+    """Fine-tune MLTC on `taxonomy_path`'s labeled rows.
+
+    With `pretrain_path`/`pretrain_steps`, first train on that (typically
+    mined, rule-labeled) taxonomy, then warm-start the fine-tune from the
+    pretrained weights with a fresh optimizer + LR schedule — the mined
+    corpus stands in for the study's unlabeled test population (reference
+    RQs/taxonomy_test2.csv is only its hand-labeled sample).
+    """
     dev = torch.device(device) if device else (
         torch.device("cuda") if torch.cuda.is_available() else
         torch.device("cpu"))
-    if taxonomy_path.endswith(".pt"):
-        full = TaxonomyDataset.from_tensor_file(taxonomy_path)
-    else:
-        df = load_taxonomy(taxonomy_path)
-        full = TaxonomyDataset.from_taxonomy(df)
-
     base = CONFIGS[model]
     cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq,
                      "dropout": dropout})
-    if full._tokens is not None and int(full._tokens.max()) >= cfg.vocab_size:
-        # prepared file was tokenized for a bigger vocab (e.g. mltc-base);
-        # fold ids into this config's hash space deterministically
-        from tosem2021_amd.models.tokenizer import N_RESERVED
-        t = full._tokens
-        big = t >= N_RESERVED
-        full._tokens = torch.where(
-            big, (t - N_RESERVED) % (cfg.vocab_size - N_RESERVED) + N_RESERVED,
-            t)
+    full = _load_dataset(taxonomy_path, cfg)
     train_ds, val_ds = full.split(val_frac=0.1, seed=seed)
     tok = CodeTokenizer(cfg.vocab_size)
     tcfg = TrainConfig(model=model, lr=lr, warmup_steps=min(50, steps // 10),
@@ -147,24 +187,26 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
         trainer.load_or_init()
 
     t0 = time.time()
-    losses = []
-    epoch = 0
-    it = iter(())
-    while trainer.step_num < steps:
-        try:
-            toks, mask, labels = next(it)
-        except StopIteration:
-            it = train_ds.batches(tok, batch, seq, device=dev, shuffle=True,
-                                  seed=seed + epoch, drop_last=True)
-            epoch += 1
-            continue
-        losses.append(trainer.step(toks, mask, labels))
-        if eval_every and trainer.step_num % eval_every == 0:
-            ev = evaluate(trainer, val_ds, tok, seq)
-            print(f"step {trainer.step_num} loss {losses[-1]:.4f} "
-                  f"strategyF1 {ev['strategy_micro_f1']:.3f} "
-                  f"propF1 {ev['property_micro_f1']:.3f} "
-                  f"methodAcc {ev['method_accuracy']:.3f}")
+    pretrain_time = 0.0
+    if pretrain_path and pretrain_steps > 0 and trainer.step_num == 0:
+        pre_ds = _load_dataset(pretrain_path, cfg)
+        pre_cfg = TrainConfig(model=model, lr=lr,
+                              warmup_steps=min(50, pretrain_steps // 10),
+                              total_steps=pretrain_steps,
+                              dtype=tcfg.dtype)
+        pre_tr = Trainer(pre_cfg, device=dev, model_cfg=cfg)
+        pre_tr.model.set_pos_weights(
+            {k: v.to(dev) for k, v in pre_ds.pos_weights().items()})
+        _run_steps(pre_tr, pre_ds, tok, batch, seq, pretrain_steps, seed,
+                   dev, eval_every, val_ds, tag="pre ")[0]
+        # warm-start: pretrained weights, fresh optimizer state + schedule
+        trainer.flat.flat.copy_(pre_tr.flat.flat)
+        trainer.flat.master.copy_(pre_tr.flat.master)
+        del pre_tr
+        pretrain_time = time.time() - t0
+
+    losses, epoch = _run_steps(trainer, train_ds, tok, batch, seq, steps,
+                               seed, dev, eval_every, val_ds)
     train_time = time.time() - t0
     ev = evaluate(trainer, val_ds, tok, seq)
     ev_lo = evaluate(trainer, val_ds, tok, seq, threshold=0.3)
@@ -180,6 +222,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
         "steps": trainer.step_num,
         "epochs": epoch,
         "train_time_s": round(train_time, 2),
+        "pretrain_time_s": round(pretrain_time, 2),
+        "pretrain_steps": pretrain_steps if pretrain_path else 0,
         "final_loss": losses[-1] if losses else None,
         "loss_first10_mean": (sum(losses[:10]) / min(len(losses), 10)
                               if losses else None),

@@ -101,7 +101,8 @@ def cmd_train(args):
     res = train_classifier(
         taxonomy_path=args.taxonomy, model=args.model, steps=args.steps,
         batch=args.batch, seq=args.seq, lr=args.lr, ckpt_dir=args.ckpt_dir,
-        resume=args.resume, dropout=args.dropout)
+        resume=args.resume, dropout=args.dropout,
+        pretrain_path=args.pretrain, pretrain_steps=args.pretrain_steps)
     print(json.dumps(res, indent=2))
 
 
@@ -167,6 +168,9 @@ def main(argv=None):
     p.add_argument("--ckpt-dir", default=None)
     p.add_argument("--resume", action="store_true")
     p.add_argument("--dropout", type=float, default=0.0)
+    p.add_argument("--pretrain", default=None,
+                   help="mined taxonomy to pretrain on before fine-tuning")
+    p.add_argument("--pretrain-steps", type=int, default=0)
     p.set_defaults(fn=cmd_train)
 
     args = ap.parse_args(argv)
