@@ -297,82 +297,85 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
 
 
 // ---------------------------------------------------------------------------
-// Flash backward, stage 1: fused recompute of P^T and dS^T.
+// Flash backward: ONE kernel recomputes P/dS and produces dS, dK and dV.
 //
-//   P^T[kv, q]  = exp(scale * S^T + mask_bias[kv] - lse[q])
-//   dP^T[kv, q] = V @ dO^T
-//   dS^T[kv, q] = scale * P^T * (dP^T - D[q]),  D = rowsum(dO * O)
+//   P[q, kv]  = exp(scale * S + mask_bias[kv] - lse[q])
+//   dP[q, kv] = dO @ V^T
+//   dS[q, kv] = scale * P * (dP - D[q]),  D = rowsum(dO * O)
+//   dV[kv, d] = sum_q P[q, kv]  * dO[q, d]   (register accumulators)
+//   dK[kv, d] = sum_q dS[q, kv] * Q[q, d]    (register accumulators)
 //
-// One MFMA pass recomputes both matmul tiles (S^T = K @ Q^T, dP^T = V @ dO^T)
-// from registers + LDS-staged Q/dO tiles; P^T and dS^T are written in the
-// [B, H, L_kv, L_q] layout (coalesced row stores), and the remaining grads
-// are three library bmms in the autograd wrapper:
-//   dV = P^T @ dO,  dK = dS^T @ Q,  dQ = (dS^T)^T @ K.
-// This replaces the unfused recompute chain (QK bmm + p_from_lse +
-// softmax_bwd) with one kernel.
+// Each wave owns one 32-row KV block: K/V fragments and the dV/dK f32
+// accumulators stay in registers for the whole kernel; the workgroup's 4
+// waves share LDS-staged Q/dO tiles (32 q rows per tile).  Per tile the
+// wave computes S and dP in the (reg=q, lane=kv) orientation — so lse/ddot
+// index by register and the mask bias is one scalar per lane — then
+// repacks P and dS into A-fragments (cvt_pk_bf16 + permlane32_swap, which
+// maps C-layout (reg=R, lane=C) to fragment [row=C][k=R]) and feeds two
+// MFMA accumulation chains against dO^T / Q^T fragments gathered with
+// ds_read_b64_tr_b16 from the already-staged tiles.  dS leaves in the
+// natural [B, H, L_q, L_kv] layout so the remaining gradient is ONE
+// non-transposed library bmm: dQ = dS @ K.  This replaces the previous
+// P^T/dS^T materialization + three batched bmms (the [512,512,64]-shaped
+// batched bmms ran at ~107 TF in hipBLASLt — see profiles/).
 //
-// Each wave owns one 32-row KV block (K/V fragments live in registers for the
-// whole kernel); the workgroup's 4 waves share the staged Q/dO tiles.
+// Mirrors reference capability only in spirit: the reference ships no
+// attention kernels (data-only replication package).
 
 extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
-flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
-                    const short* __restrict__ v, const short* __restrict__ dout,
-                    const float* __restrict__ mask,
-                    const float* __restrict__ lse,
-                    const float* __restrict__ ddot,
-                    short* __restrict__ p_t, short* __restrict__ ds_t,
-                    int B, int H, int L, float scale) {
-  // Each wave owns TWO 32-row KV blocks (K/V fragments resident for both):
-  // the second block's MFMA/elementwise chains overlap the first block's
-  // serial work.  2 waves/SIMD by registers; the workgroup covers 256 kv
-  // rows.  Mask bias is staged in LDS (registers are the scarce resource).
+flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
+                       const short* __restrict__ v,
+                       const short* __restrict__ dout,
+                       const float* __restrict__ mask,
+                       const float* __restrict__ lse,
+                       const float* __restrict__ ddot,
+                       short* __restrict__ ds, short* __restrict__ dk,
+                       short* __restrict__ dv,
+                       int B, int H, int L, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = (short*)smem;                        // swizzled [32][64]
   short* do_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
-  float* mb_lds = (float*)(smem + 2 * K_LDS_BYTES);   // [256] kv mask bias
+  float* lse_lds = (float*)(smem + 2 * K_LDS_BYTES);  // [32]
+  float* dd_lds = lse_lds + 32;                       // [32]
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
-  const int col = lane & 31;        // q column of this lane
+  const int col = lane & 31;
   const int half = lane >> 5;
 
-  const int rows_per_wg = 2 * FA_QWG;                 // 256 kv rows per WG
-  const int n_kvblocks = (L + rows_per_wg - 1) / rows_per_wg;
+  const int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;   // 128 kv rows per WG
   int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_kvblocks);
   int bh = bid / n_kvblocks;
   int kb = bid % n_kvblocks;
   const int b = bh / H;
   const long bh_off = (long)bh * L * FA_DH;
   const long bh_sq = (long)bh * L * L;
-  const int kv_wg = kb * rows_per_wg;
-  const int kv_baseA = kv_wg + wid * FA_KVB;          // wave's first kv block
-  const int kv_baseB = kv_baseA + FA_QWG;             // second, 128 later
+  const int kv_base = kb * FA_QWG + wid * FA_KVB;     // this wave's 32 kv rows
+  const int my_kv = kv_base + col;
+  const bool kv_valid = my_kv < L;                    // L%32==0: whole block
   const float* mrow = mask ? mask + (long)b * L : nullptr;
+  const float mb = mrow ? mrow[min(my_kv, L - 1)] : 0.f;
 
-  // stage the WG's 256 mask-bias values once
-  if (tid < rows_per_wg)
-    mb_lds[tid] = (mrow && kv_wg + tid < L) ? mrow[kv_wg + tid] : 0.f;
-
-  // K and V fragments for both kv blocks (resident all kernel)
-  short8_t kfA[4], vfA[4], kfB[4], vfB[4];
+  // K and V fragments for this wave's kv block (resident all kernel)
+  short8_t kf[4], vf[4];
   {
-    const int mkA = min(kv_baseA + col, L - 1);
-    const int mkB = min(kv_baseB + col, L - 1);
-    const short* krA = k + bh_off + (long)mkA * FA_DH;
-    const short* vrA = v + bh_off + (long)mkA * FA_DH;
-    const short* krB = k + bh_off + (long)mkB * FA_DH;
-    const short* vrB = v + bh_off + (long)mkB * FA_DH;
+    const short* kr = k + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
+    const short* vr = v + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
-      kfA[c] = *(const short8_t*)(krA + c * 16 + half * 8);
-      vfA[c] = *(const short8_t*)(vrA + c * 16 + half * 8);
-      kfB[c] = *(const short8_t*)(krB + c * 16 + half * 8);
-      vfB[c] = *(const short8_t*)(vrB + c * 16 + half * 8);
+      kf[c] = *(const short8_t*)(kr + c * 16 + half * 8);
+      vf[c] = *(const short8_t*)(vr + c * 16 + half * 8);
     }
   }
+  f32x16 dv_acc[2], dk_acc[2];
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    dv_acc[t] = (f32x16)(0.f);
+    dk_acc[t] = (f32x16)(0.f);
+  }
 
-  const int n_q = L / FA_KVB;
+  const int n_q = L / 32;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
   short8_t qv8 = *(const short8_t*)(q + bh_off + (long)srow * FA_DH +
                                     (sc8 >> 1));
@@ -384,6 +387,10 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
     {
       *(short8_t*)((char*)q_lds + srow * 128 + kswz(srow, sc8)) = qv8;
       *(short8_t*)((char*)do_lds + srow * 128 + kswz(srow, sc8)) = dv8;
+      if (tid < 32) {
+        lse_lds[tid] = lse[(long)bh * L + q0 + tid];
+        dd_lds[tid] = ddot[(long)bh * L + q0 + tid];
+      }
     }
     __syncthreads();
     if (qt + 1 < n_q) {
@@ -393,9 +400,9 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                (sc8 >> 1));
     }
 
-    // S^T = K @ Q^T and dP^T = V @ dO^T for BOTH kv blocks (16 MFMAs)
-    f32x16 sA = (f32x16)(0.f), dpA = (f32x16)(0.f);
-    f32x16 sB = (f32x16)(0.f), dpB = (f32x16)(0.f);
+    // S[q, kv] and dP[q, kv]: reg=q rows, lane=kv cols
+    f32x16 s_acc = (f32x16)(0.f);
+    f32x16 dp_acc = (f32x16)(0.f);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
@@ -404,66 +411,168 @@ flash_bwd_ds_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                           kswz(col, byte_off));
       short8_t dofrag = *(const short8_t*)((char*)do_lds + col * 128 +
                                            kswz(col, byte_off));
-      sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[c], qfrag, sA, 0, 0, 0);
-      dpA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[c], dofrag, dpA,
-                                                    0, 0, 0);
-      sB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfB[c], qfrag, sB, 0, 0, 0);
-      dpB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfB[c], dofrag, dpB,
-                                                    0, 0, 0);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag, kf[c], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag, vf[c], dp_acc,
+                                                       0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // elementwise + store, block A then block B (independent chains)
-    const int my_q = q0 + col;
-    const float l_q = lse[(long)bh * L + my_q];
-    const float d_q = ddot[(long)bh * L + my_q];
+    // elementwise (lse/ddot by register row, mask bias one scalar per lane)
+    float pv[16], gv[16];
 #pragma unroll
-    for (int blk = 0; blk < 2; ++blk) {
-      const f32x16& s_acc = blk ? sB : sA;
-      const f32x16& dp_acc = blk ? dpB : dpA;
-      const int kv_base = blk ? kv_baseB : kv_baseA;
-      const int mb0 = kv_base - kv_wg;
-      short8_t p_pack[2], ds_pack[2];
+    for (int r = 0; r < 16; ++r) {
+      int ql = (r & 3) + 8 * (r >> 2) + 4 * half;
+      pv[r] = __expf(s_acc[r] * scale + mb - lse_lds[ql]);
+      gv[r] = scale * pv[r] * (dp_acc[r] - dd_lds[ql]);
+    }
+    // dS out, natural [q, kv] rows (2 B per lane, lanes contiguous in kv)
+    if (kv_valid) {
 #pragma unroll
-      for (int rr = 0; rr < 16; rr += 2) {
-        int kl0 = (rr & 3) + 8 * (rr >> 2) + 4 * half;
-        float p0 = __expf(s_acc[rr] * scale + mb_lds[mb0 + kl0] - l_q);
-        float p1 = __expf(s_acc[rr + 1] * scale + mb_lds[mb0 + kl0 + 1] - l_q);
-        float g0 = scale * p0 * (dp_acc[rr] - d_q);
-        float g1 = scale * p1 * (dp_acc[rr + 1] - d_q);
-        int slot = rr >> 1;  // 8 packed pairs
-        p_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(p0);
-        p_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(p1);
-        ds_pack[slot >> 2][2 * (slot & 3)] = f32_to_bf16(g0);
-        ds_pack[slot >> 2][2 * (slot & 3) + 1] = f32_to_bf16(g1);
+      for (int r = 0; r < 16; ++r) {
+        int ql = (r & 3) + 8 * (r >> 2) + 4 * half;
+        ds[bh_sq + (long)(q0 + ql) * L + my_kv] = f32_to_bf16(gv[r]);
       }
-      // store: rows kv = kv_base + crow(2*slot..), cols q0 + col (2 B/lane)
+    }
+
+    // repack P and dS to A-fragments [row=kv][k=q]
+    short8_t pf[2], gf[2];
 #pragma unroll
-      for (int rr = 0; rr < 16; ++rr) {
-        int kvl = (rr & 3) + 8 * (rr >> 2) + 4 * half;
-        if (kv_base + kvl >= L) continue;  // tail kv-block: no OOB rows
-        long off = bh_sq + (long)(kv_base + kvl) * L + q0 + col;
-        int slot = rr >> 1;
-        p_t[off] = p_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
-        ds_t[off] = ds_pack[slot >> 2][2 * (slot & 3) + (rr & 1)];
+    for (int c = 0; c < 2; ++c) {
+      typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+      uint4_t up, ug;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int r0 = c * 8 + 2 * i;
+        int r1 = c * 8 + 4 + 2 * i;
+        unsigned lo, hi;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
+            : "v"(pv[r0]), "v"(pv[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
+            : "v"(pv[r1]), "v"(pv[r1 + 1]));
+        auto sw = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
+        up[i] = sw[0]; up[i + 2] = sw[1];
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(lo)
+            : "v"(gv[r0]), "v"(gv[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hi)
+            : "v"(gv[r1]), "v"(gv[r1 + 1]));
+        auto sw2 = __builtin_amdgcn_permlane32_swap(lo, hi, false, false);
+        ug[i] = sw2[0]; ug[i + 2] = sw2[1];
       }
+      pf[c] = __builtin_bit_cast(short8_t, up);
+      gf[c] = __builtin_bit_cast(short8_t, ug);
+    }
+
+    // dO^T and Q^T B-fragments [row=d][k=q] via lane-grid transpose reads,
+    // then the two accumulation chains (reg=kv rows, lane=d cols)
+    {
+      typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
+      const unsigned dobase = (unsigned)(unsigned long)(char*)do_lds;
+      const unsigned qbase = (unsigned)(unsigned long)(char*)q_lds;
+      const int q_mate = (lane >> 2) & 3;
+      const int d_lane = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+      unsigned ad[8], aq[8];
+#pragma unroll
+      for (int c = 0; c < 2; ++c)
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
+#pragma unroll
+          for (int t = 0; t < 2; ++t) {
+            int qq = 16 * c + 8 * half + 4 * rr + q_mate;
+            int dcol = (32 * t + d_lane) * 2;
+            int off = qq * 128 + (dcol ^ ((qq & 7) << 4));
+            ad[c * 4 + rr * 2 + t] = dobase + off;
+            aq[c * 4 + rr * 2 + t] = qbase + off;
+          }
+      uint2_t rd[8], rq[8];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8\n\t"
+          "ds_read_b64_tr_b16 %1, %9\n\t"
+          "ds_read_b64_tr_b16 %2, %10\n\t"
+          "ds_read_b64_tr_b16 %3, %11\n\t"
+          "ds_read_b64_tr_b16 %4, %12\n\t"
+          "ds_read_b64_tr_b16 %5, %13\n\t"
+          "ds_read_b64_tr_b16 %6, %14\n\t"
+          "ds_read_b64_tr_b16 %7, %15\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(rd[0]), "=&v"(rd[1]), "=&v"(rd[2]), "=&v"(rd[3]),
+            "=&v"(rd[4]), "=&v"(rd[5]), "=&v"(rd[6]), "=&v"(rd[7])
+          : "v"(ad[0]), "v"(ad[1]), "v"(ad[2]), "v"(ad[3]), "v"(ad[4]),
+            "v"(ad[5]), "v"(ad[6]), "v"(ad[7])
+          : "memory");
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8\n\t"
+          "ds_read_b64_tr_b16 %1, %9\n\t"
+          "ds_read_b64_tr_b16 %2, %10\n\t"
+          "ds_read_b64_tr_b16 %3, %11\n\t"
+          "ds_read_b64_tr_b16 %4, %12\n\t"
+          "ds_read_b64_tr_b16 %5, %13\n\t"
+          "ds_read_b64_tr_b16 %6, %14\n\t"
+          "ds_read_b64_tr_b16 %7, %15\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(rq[0]), "=&v"(rq[1]), "=&v"(rq[2]), "=&v"(rq[3]),
+            "=&v"(rq[4]), "=&v"(rq[5]), "=&v"(rq[6]), "=&v"(rq[7])
+          : "v"(aq[0]), "v"(aq[1]), "v"(aq[2]), "v"(aq[3]), "v"(aq[4]),
+            "v"(aq[5]), "v"(aq[6]), "v"(aq[7])
+          : "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+          uint4_t wd, wq;
+          wd[0] = rd[c * 4 + 0 * 2 + t][0];
+          wd[1] = rd[c * 4 + 0 * 2 + t][1];
+          wd[2] = rd[c * 4 + 1 * 2 + t][0];
+          wd[3] = rd[c * 4 + 1 * 2 + t][1];
+          wq[0] = rq[c * 4 + 0 * 2 + t][0];
+          wq[1] = rq[c * 4 + 0 * 2 + t][1];
+          wq[2] = rq[c * 4 + 1 * 2 + t][0];
+          wq[3] = rq[c * 4 + 1 * 2 + t][1];
+          short8_t dof = __builtin_bit_cast(short8_t, wd);
+          short8_t qf = __builtin_bit_cast(short8_t, wq);
+          dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[c], dof,
+                                                              dv_acc[t],
+                                                              0, 0, 0);
+          dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(gf[c], qf,
+                                                              dk_acc[t],
+                                                              0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+  // epilogue: dV/dK rows of this wave's kv block (reg=kv row, lane=d col)
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (kv_base + kvl >= L) continue;
+      long off = bh_off + (long)(kv_base + kvl) * FA_DH + 32 * t + col;
+      dv[off] = f32_to_bf16(dv_acc[t][r]);
+      dk[off] = f32_to_bf16(dk_acc[t][r]);
     }
   }
 }
 
-extern "C" hipError_t flash_bwd_ds_launch(
+extern "C" hipError_t flash_bwd_fused_launch(
     const void* q, const void* k, const void* v, const void* dout,
-    const void* mask, const void* lse, const void* ddot, void* p_t,
-    void* ds_t, int B, int H, int L, float scale, hipStream_t stream) {
-  int n_kvblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
+    const void* mask, const void* lse, const void* ddot, void* ds, void* dk,
+    void* dv, int B, int H, int L, float scale, hipStream_t stream) {
+  int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;
   dim3 grid(B * H * n_kvblocks);
-  size_t shm = 2 * K_LDS_BYTES + 2 * FA_QWG * sizeof(float);
-  flash_bwd_ds_kernel<<<grid, FA_BLOCK, shm, stream>>>(
+  size_t shm = 2 * K_LDS_BYTES + 64 * sizeof(float);
+  flash_bwd_fused_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
-      (const float*)mask, (const float*)lse, (const float*)ddot, (short*)p_t,
-      (short*)ds_t, B, H, L, scale);
+      (const float*)mask, (const float*)lse, (const float*)ddot, (short*)ds,
+      (short*)dk, (short*)dv, B, H, L, scale);
   return hipGetLastError();
 }
+
 
 
 // D = rowsum(dO * O) per (b, h, q) row — one wave per 4 rows (dh = 64).

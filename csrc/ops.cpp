@@ -39,10 +39,10 @@ hipError_t flash_fwd_launch(const void*, const void*, const void*, const void*,
                             void*, void*, int, int, int, float, hipStream_t);
 hipError_t p_from_lse_launch(const void*, const void*, const void*, void*,
                              long, int, int, float, int, hipStream_t);
-hipError_t flash_bwd_ds_launch(const void*, const void*, const void*,
-                               const void*, const void*, const void*,
-                               const void*, void*, void*, int, int, int,
-                               float, hipStream_t);
+hipError_t flash_bwd_fused_launch(const void*, const void*, const void*,
+                                  const void*, const void*, const void*,
+                                  const void*, void*, void*, void*, int, int,
+                                  int, float, hipStream_t);
 hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
 hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
                                   int, int, hipStream_t);
@@ -348,11 +348,11 @@ torch::Tensor fa_dot(torch::Tensor dout, torch::Tensor o) {
   return d.view({dout.size(0), dout.size(1), dout.size(2)});
 }
 
-std::vector<torch::Tensor> flash_bwd_ds(torch::Tensor q, torch::Tensor k,
-                                        torch::Tensor v, torch::Tensor dout,
-                                        c10::optional<torch::Tensor> mask,
-                                        torch::Tensor lse, torch::Tensor ddot,
-                                        double scale) {
+std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, torch::Tensor dout,
+                                           c10::optional<torch::Tensor> mask,
+                                           torch::Tensor lse,
+                                           torch::Tensor ddot, double scale) {
   check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
   check_bf16(dout, "dout"); check_f32(lse, "lse"); check_f32(ddot, "ddot");
   const long B = q.size(0), H = q.size(1), L = q.size(2);
@@ -362,14 +362,16 @@ std::vector<torch::Tensor> flash_bwd_ds(torch::Tensor q, torch::Tensor k,
     check_f32(*mask, "mask");
     mptr = mask->data_ptr();
   }
-  auto p_t = torch::empty({B, H, L, L}, q.options());
-  auto ds_t = torch::empty({B, H, L, L}, q.options());
-  CHECK_HIP(flash_bwd_ds_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                                dout.data_ptr(), mptr, lse.data_ptr(),
-                                ddot.data_ptr(), p_t.data_ptr(),
-                                ds_t.data_ptr(), (int)B, (int)H, (int)L,
-                                (float)scale, cur_stream()));
-  return {p_t, ds_t};
+  auto ds = torch::empty({B, H, L, L}, q.options());
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  CHECK_HIP(flash_bwd_fused_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                                   dout.data_ptr(), mptr, lse.data_ptr(),
+                                   ddot.data_ptr(), ds.data_ptr(),
+                                   dk.data_ptr(), dv.data_ptr(), (int)B,
+                                   (int)H, (int)L, (float)scale,
+                                   cur_stream()));
+  return {ds, dk, dv};
 }
 
 std::vector<torch::Tensor> masked_pool_fwd(torch::Tensor x,
@@ -420,7 +422,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("masked_pool_fwd", &masked_pool_fwd, "masked mean-pool fwd");
   m.def("masked_pool_bwd", &masked_pool_bwd, "masked mean-pool bwd");
   m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
-  m.def("flash_bwd_ds", &flash_bwd_ds, "fused flash bwd: P^T and dS^T tiles");
+  m.def("flash_bwd_fused", &flash_bwd_fused,
+        "fused flash bwd: dS + register-accumulated dK/dV");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
   m.def("p_from_lse", &p_from_lse, "probabilities from saved logsumexp");
   m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");

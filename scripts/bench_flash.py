@@ -33,14 +33,25 @@ def main():
     do = torch.randn_like(q)
     ddot = ext.fa_dot(do, o)
     for _ in range(3):
-        p_t, ds_t = ext.flash_bwd_ds(q, k, v, do, mask, lse, ddot, scale)
+        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        p_t, ds_t = ext.flash_bwd_ds(q, k, v, do, mask, lse, ddot, scale)
+        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    print(f"flash_bwd_ds {dt*1e6:.1f} us/call  {2*flops/dt/1e12:.1f} TF-eq")
+    print(f"flash_bwd_fused {dt*1e6:.1f} us/call  {4*flops/dt/1e12:.1f} TF-eq")
+    # whole backward chain incl. the one remaining bmm (dq)
+    for _ in range(3):
+        dq = torch.matmul(ds, k)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
+        dq = torch.matmul(ds, k)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    print(f"bwd chain (fused + dq bmm) {dt*1e6:.1f} us")
 
 
 if __name__ == "__main__":

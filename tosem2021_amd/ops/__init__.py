@@ -221,7 +221,7 @@ def out_repack(x):
 
 class _FlashAttentionFn(torch.autograd.Function):
     """Flash attention: MFMA forward (csrc/flash_attn.hip, O(L) memory, saves
-    logsumexp), recompute backward (bmm + p_from_lse + softmax_bwd)."""
+    logsumexp), fused recompute backward (dS + in-kernel dK/dV)."""
 
     @staticmethod
     def forward(ctx, q, k, v, mask, scale):
@@ -240,13 +240,12 @@ class _FlashAttentionFn(torch.autograd.Function):
         mask, scale = ctx.mask, ctx.scale
         do = do.contiguous()
         if q.is_cuda:
-            # fused recompute: one MFMA kernel emits P^T and dS^T directly
+            # fully-fused recompute: one MFMA kernel emits dS and
+            # register-accumulated dK/dV; dQ is the one remaining bmm
             ddot = hip_ops().fa_dot(do, o)
-            p_t, ds_t = hip_ops().flash_bwd_ds(q, k, v, do, mask, lse, ddot,
-                                               scale)
-            dv = torch.matmul(p_t, do)  # [kv,q] @ [q,d]
-            dk = torch.matmul(ds_t, q)
-            dq = torch.matmul(ds_t.transpose(-1, -2), k)
+            ds, dk, dv = hip_ops().flash_bwd_fused(q, k, v, do, mask, lse,
+                                                   ddot, scale)
+            dq = torch.matmul(ds, k)  # [q,kv] @ [kv,d]
             return dq, dk, dv, None, None
         s = torch.matmul(q, k.transpose(-1, -2))
         p = reference.p_from_lse(s, mask, lse, scale)
