@@ -140,6 +140,198 @@ ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
   }
 }
 
+}  // extern "C" (templates below need C++ linkage)
+
+// Template-specialized fast path: exact unroll for D = PKTS*512 (no
+// MAX_PKT-sized register waste), gamma/beta hoisted out of the row loop,
+// and the NEXT row's packets prefetched before the wave-reduction chains so
+// the loads overlap the two log2(64) shuffle reductions.  One wave per row.
+template <int PKTS, bool HASRES>
+__global__ void __launch_bounds__(BLOCK)
+ln_fwd_t(const short* __restrict__ x, const short* __restrict__ res,
+         const short* __restrict__ gamma, const short* __restrict__ beta,
+         short* __restrict__ y, short* __restrict__ s_out,
+         float* __restrict__ mean_out, float* __restrict__ rstd_out,
+         int N, int D, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rstride = gridDim.x * WAVES_PER_BLOCK;
+  int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (row >= N) return;
+  float gv[PKTS * 8], bv[PKTS * 8];
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    int base = (p * WAVE + lane) * 8;
+    short8_t g8 = *(const short8_t*)(gamma + base);
+    short8_t b8 = *(const short8_t*)(beta + base);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      gv[p * 8 + j] = bf16_to_f32(g8[j]);
+      bv[p * 8 + j] = bf16_to_f32(b8[j]);
+    }
+  }
+  short8_t vx[PKTS], vr[PKTS];
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    int base = (p * WAVE + lane) * 8;
+    vx[p] = *(const short8_t*)(x + (long)row * D + base);
+    if (HASRES) vr[p] = *(const short8_t*)(res + (long)row * D + base);
+  }
+  while (true) {
+    const int next = row + rstride;
+    float vals[PKTS * 8];
+    float s = 0.f;
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      short8_t so;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32(vx[p][j]);
+        if (HASRES) {
+          // bf16-rounded sum so s_out, the saved x for bwd, and the stats
+          // all see the SAME residual-stream value
+          short sum_b = f32_to_bf16(f + bf16_to_f32(vr[p][j]));
+          f = bf16_to_f32(sum_b);
+          so[j] = sum_b;
+        }
+        vals[p * 8 + j] = f;
+        s += f;
+      }
+      if (HASRES)
+        *(short8_t*)(s_out + (long)row * D + (p * WAVE + lane) * 8) = so;
+    }
+    if (next < N) {
+#pragma unroll
+      for (int p = 0; p < PKTS; ++p) {
+        int base = (p * WAVE + lane) * 8;
+        vx[p] = *(const short8_t*)(x + (long)next * D + base);
+        if (HASRES) vr[p] = *(const short8_t*)(res + (long)next * D + base);
+      }
+    }
+    float mean = wave_sum(s) / (float)D;
+    float var = 0.f;
+#pragma unroll
+    for (int k = 0; k < PKTS * 8; ++k) {
+      float d = vals[k] - mean;
+      var += d * d;
+    }
+    var = wave_sum(var) / (float)D;
+    float rstd = rsqrtf(var + eps);
+    if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      short8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = p * 8 + j;
+        o[j] = f32_to_bf16((vals[k] - mean) * rstd * gv[k] + bv[k]);
+      }
+      *(short8_t*)(y + (long)row * D + (p * WAVE + lane) * 8) = o;
+    }
+    if (next >= N) break;
+    row = next;
+  }
+}
+
+template <int PKTS, bool HASDE>
+__global__ void __launch_bounds__(BLOCK)
+ln_bwd_t(const short* __restrict__ dy, const short* __restrict__ x,
+         const short* __restrict__ gamma, const float* __restrict__ mean_in,
+         const float* __restrict__ rstd_in,
+         const short* __restrict__ ds_extra, short* __restrict__ dx,
+         float* __restrict__ ws_dgamma, float* __restrict__ ws_dbeta,
+         int N, int D) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sg = (float*)smem;
+  float* sb = sg + D;
+  for (int i = threadIdx.x; i < D; i += BLOCK) { sg[i] = 0.f; sb[i] = 0.f; }
+  __syncthreads();
+  float gv[PKTS * 8];
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    short8_t g8 = *(const short8_t*)(gamma + (p * WAVE + lane) * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) gv[p * 8 + j] = bf16_to_f32(g8[j]);
+  }
+  float dg_acc[PKTS * 8] = {0.f}, db_acc[PKTS * 8] = {0.f};
+  const int rstride = gridDim.x * WAVES_PER_BLOCK;
+  int row = blockIdx.x * WAVES_PER_BLOCK + wid;
+  const bool active = row < N;
+  short8_t vd[PKTS], vxp[PKTS], ve[PKTS];
+  if (active) {
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      int base = (p * WAVE + lane) * 8;
+      vd[p] = *(const short8_t*)(dy + (long)row * D + base);
+      vxp[p] = *(const short8_t*)(x + (long)row * D + base);
+      if (HASDE) ve[p] = *(const short8_t*)(ds_extra + (long)row * D + base);
+    }
+  }
+  while (active) {
+    const int next = row + rstride;
+    const float mean = mean_in[row], rstd = rstd_in[row];
+    float xh[PKTS * 8], dyg[PKTS * 8], ev[PKTS * 8];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = p * 8 + j;
+        float d = bf16_to_f32(vd[p][j]);
+        float h = (bf16_to_f32(vxp[p][j]) - mean) * rstd;
+        float g = d * gv[k];
+        if (HASDE) ev[k] = bf16_to_f32(ve[p][j]);
+        xh[k] = h; dyg[k] = g;
+        dg_acc[k] += d * h;
+        db_acc[k] += d;
+        s1 += g; s2 += g * h;
+      }
+    }
+    if (next < N) {
+#pragma unroll
+      for (int p = 0; p < PKTS; ++p) {
+        int base = (p * WAVE + lane) * 8;
+        vd[p] = *(const short8_t*)(dy + (long)next * D + base);
+        vxp[p] = *(const short8_t*)(x + (long)next * D + base);
+        if (HASDE) ve[p] = *(const short8_t*)(ds_extra + (long)next * D + base);
+      }
+    }
+    s1 = wave_sum(s1) / (float)D;
+    s2 = wave_sum(s2) / (float)D;
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      short8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = p * 8 + j;
+        float d_ = rstd * (dyg[k] - s1 - xh[k] * s2);
+        if (HASDE) d_ += ev[k];
+        o[j] = f32_to_bf16(d_);
+      }
+      *(short8_t*)(dx + (long)row * D + (p * WAVE + lane) * 8) = o;
+    }
+    if (next >= N) break;
+    row = next;
+  }
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    int base = (p * WAVE + lane) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sg[base + j], dg_acc[p * 8 + j]);
+      atomicAdd(&sb[base + j], db_acc[p * 8 + j]);
+    }
+  }
+  __syncthreads();
+  float* og = ws_dgamma + (long)blockIdx.x * D;
+  float* ob = ws_dbeta + (long)blockIdx.x * D;
+  for (int i = threadIdx.x; i < D; i += BLOCK) { og[i] = sg[i]; ob[i] = sb[i]; }
+}
+
+extern "C" {
+
 // dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg * xhat)),  dyg = dy * gamma
 // Per-block dgamma/dbeta partials go to ws_dgamma/ws_dbeta [gridDim.x, D]
 // (deterministic two-stage column reduction, no atomics).
@@ -300,10 +492,29 @@ hipError_t ln_fwd_launch(const void* x, const void* res, const void* gamma,
                          const void* beta, void* y, void* s_out, void* mean,
                          void* rstd, int N, int D, float eps, int grid,
                          hipStream_t stream) {
-  ln_fwd_kernel<<<grid, BLOCK, 0, stream>>>(
-      (const short*)x, (const short*)res, (const short*)gamma,
-      (const short*)beta, (short*)y, (short*)s_out, (float*)mean,
-      (float*)rstd, N, D, eps);
+#define LNF_T(P)                                                              \
+  do {                                                                        \
+    if (res)                                                                  \
+      ln_fwd_t<P, true><<<grid, BLOCK, 0, stream>>>(                          \
+          (const short*)x, (const short*)res, (const short*)gamma,            \
+          (const short*)beta, (short*)y, (short*)s_out, (float*)mean,         \
+          (float*)rstd, N, D, eps);                                           \
+    else                                                                      \
+      ln_fwd_t<P, false><<<grid, BLOCK, 0, stream>>>(                         \
+          (const short*)x, nullptr, (const short*)gamma, (const short*)beta,  \
+          (short*)y, (short*)s_out, (float*)mean, (float*)rstd, N, D, eps);   \
+  } while (0)
+  if (D == 512) LNF_T(1);
+  else if (D == 1024) LNF_T(2);
+  else if (D == 2048) LNF_T(4);
+  // PKTS=8 (D=4096) allocates 256 VGPRs (1 wave/SIMD) — the general kernel
+  // is the better trade there
+  else
+    ln_fwd_kernel<<<grid, BLOCK, 0, stream>>>(
+        (const short*)x, (const short*)res, (const short*)gamma,
+        (const short*)beta, (short*)y, (short*)s_out, (float*)mean,
+        (float*)rstd, N, D, eps);
+#undef LNF_T
   return hipGetLastError();
 }
 
@@ -313,14 +524,31 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
                          void* ws_dgamma, void* ws_dbeta, int N, int D,
                          int grid, hipStream_t stream) {
   size_t shm = (size_t)D * 2 * sizeof(float);
-  ln_bwd_kernel<<<grid, BLOCK, shm, stream>>>(
-      (const short*)dy, (const short*)x, (const short*)gamma,
-      (const float*)mean, (const float*)rstd, (const short*)ds_extra,
-      (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);
+#define LNB_T(P)                                                              \
+  do {                                                                        \
+    if (ds_extra)                                                             \
+      ln_bwd_t<P, true><<<grid, BLOCK, shm, stream>>>(                        \
+          (const short*)dy, (const short*)x, (const short*)gamma,             \
+          (const float*)mean, (const float*)rstd, (const short*)ds_extra,     \
+          (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);             \
+    else                                                                      \
+      ln_bwd_t<P, false><<<grid, BLOCK, shm, stream>>>(                       \
+          (const short*)dy, (const short*)x, (const short*)gamma,             \
+          (const float*)mean, (const float*)rstd, nullptr, (short*)dx,        \
+          (float*)ws_dgamma, (float*)ws_dbeta, N, D);                         \
+  } while (0)
+  if (D == 512) LNB_T(1);
+  else if (D == 1024) LNB_T(2);
+  else
+    ln_bwd_kernel<<<grid, BLOCK, shm, stream>>>(
+        (const short*)dy, (const short*)x, (const short*)gamma,
+        (const float*)mean, (const float*)rstd, (const short*)ds_extra,
+        (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);
+#undef LNB_T
   return hipGetLastError();
 }
 
-#define COLSUM_SPLITS 16
+#define COLSUM_SPLITS 64
 hipError_t colsum_launch(const void* ws, void* scratch, void* out, int R,
                          int D, hipStream_t stream) {
   dim3 grid1((D / 4 + 255) / 256, 1);
