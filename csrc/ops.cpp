@@ -51,6 +51,10 @@ hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
 hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
 }
 
+torch::Tensor lt_linear_gelu_bias(torch::Tensor, torch::Tensor,
+                                  torch::Tensor);
+bool lt_probe_epilogue(long, long, long, long);
+
 namespace {
 
 #define CHECK_HIP(call)                                                   \
@@ -435,6 +439,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (bf16, gfx950)");
   m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU fwd");
+  m.def("lt_linear_gelu_bias", &lt_linear_gelu_bias,
+        "hipBLASLt GEMM with fused GELU_BIAS epilogue (inference)");
+  m.def("lt_probe_epilogue", &lt_probe_epilogue,
+        "does this hipBLASLt have kernels for (m, n, k, epilogue)?");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU bwd");
   m.def("softmax_fwd", &softmax_fwd, "fused scaled masked softmax fwd");
   m.def("softmax_bwd", &softmax_bwd, "fused softmax bwd");

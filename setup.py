@@ -24,6 +24,7 @@ ext = CUDAExtension(
     name="tosem2021_amd._hip_ops",
     sources=[
         "csrc/ops.cpp",
+        "csrc/lt_gemm.cpp",
         "csrc/layernorm.hip",
         "csrc/bias_gelu.hip",
         "csrc/softmax.hip",
@@ -32,6 +33,7 @@ ext = CUDAExtension(
         "csrc/flash_attn.hip",
         "csrc/pool.hip",
     ],
+    libraries=["hipblaslt"],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

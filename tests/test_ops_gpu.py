@@ -329,3 +329,38 @@ def test_masked_pool_fully_masked_row():
     assert float(counts[1]) == 1.0  # clamped, no div-by-zero
     assert torch.all(pooled[1].float() == 0)
     assert not torch.isnan(pooled.float()).any()
+
+
+@pytest.mark.gpu
+def test_lt_linear_gelu_bias_vs_reference():
+    """hipBLASLt GELU_BIAS epilogue (inference FFN-up) vs fp32 torch; also
+    pins which epilogues this hipBLASLt build provides (csrc/lt_gemm.cpp)."""
+    ext = ops.hip_ops()
+    assert ext.lt_probe_epilogue(1024, 512, 256, 36)      # GELU_BIAS
+    torch.manual_seed(21)
+    N, D, F = 512, 256, 1024
+    x = _bf16(torch.randn(N, D))
+    w1 = _bf16(torch.randn(F, D) * 0.05)
+    b1 = _bf16(torch.randn(F) * 0.1)
+    a = ext.lt_linear_gelu_bias(x, w1, b1)
+    h_ref = x.float() @ w1.float().T + b1.float()
+    a_ref = torch.nn.functional.gelu(h_ref, approximate="tanh")
+    a_ref_erf = torch.nn.functional.gelu(h_ref)
+    d_tanh = (a.float() - a_ref).abs().max()
+    d_erf = (a.float() - a_ref_erf).abs().max()
+    assert min(d_tanh, d_erf) < 3e-2, (d_tanh, d_erf)
+
+
+@pytest.mark.gpu
+def test_ffn_inference_path_matches_training_path():
+    """FFN.forward under no_grad (GELU_BIAS epilogue) vs the grad-mode
+    kernel path on the same weights."""
+    from tosem2021_amd.models.classifier import CONFIGS, FFN
+    torch.manual_seed(22)
+    ffn = FFN(CONFIGS["mltc-tiny"]).to(torch.bfloat16).cuda()
+    x = _bf16(torch.randn(4, 64, CONFIGS["mltc-tiny"].d_model))
+    with torch.no_grad():
+        y_inf = ffn(x)
+    y_train = ffn(x.requires_grad_())
+    assert torch.allclose(y_inf.float(), y_train.float(), atol=3e-2,
+                          rtol=3e-2), (y_inf.float() - y_train.float()).abs().max()

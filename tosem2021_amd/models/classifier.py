@@ -118,6 +118,12 @@ class FFN(nn.Module):
         self.down = nn.Linear(cfg.d_ff, cfg.d_model, bias=True)
 
     def forward(self, x):
+        if x.is_cuda and not torch.is_grad_enabled():
+            # inference: bias+GeLU fused into the up GEMM's epilogue
+            B, L, D = x.shape
+            a = ops.lt_linear_gelu_bias(x.reshape(-1, D).contiguous(),
+                                        self.up.weight, self.up_bias)
+            return self.down(a.view(B, L, -1))
         h = self.up(x)
         h = ops.fused_bias_gelu(h, self.up_bias)
         return self.down(h)

@@ -219,6 +219,13 @@ def out_repack(x):
     return _OutRepackFn.apply(x.contiguous())
 
 
+def lt_linear_gelu_bias(x, w1, b1):
+    """GELU(x @ w1^T + b1) in one hipBLASLt GEMM (GELU_BIAS epilogue) —
+    inference only: this hipBLASLt has no aux epilogues (no pre-activation
+    out), so training uses csrc/bias_gelu.hip instead."""
+    return hip_ops().lt_linear_gelu_bias(x, w1, b1)
+
+
 class _FlashAttentionFn(torch.autograd.Function):
     """Flash attention: MFMA forward (csrc/flash_attn.hip, O(L) memory, saves
     logsumexp), fused recompute backward (dS + in-kernel dK/dV)."""
