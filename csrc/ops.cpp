@@ -44,6 +44,8 @@ hipError_t flash_bwd_fused_launch(const void*, const void*, const void*,
                                   const void*, void*, void*, void*, int, int,
                                   int, float, hipStream_t);
 hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
+hipError_t flash_dq_launch(const void*, const void*, void*, int, int, int,
+                           hipStream_t);
 hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
                                   int, int, hipStream_t);
 hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
@@ -380,6 +382,17 @@ std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
   return {ds, dk, dv};
 }
 
+torch::Tensor flash_dq(torch::Tensor ds, torch::Tensor k) {
+  check_bf16(ds, "ds"); check_bf16(k, "k");
+  const long B = k.size(0), H = k.size(1), L = k.size(2);
+  TORCH_CHECK(k.size(3) == 64 && L % 32 == 0, "dh=64 and L%32==0 required");
+  TORCH_CHECK(ds.size(2) == L && ds.size(3) == L, "ds must be [B,H,L,L]");
+  auto dq = torch::empty_like(k);
+  CHECK_HIP(flash_dq_launch(ds.data_ptr(), k.data_ptr(), dq.data_ptr(),
+                            (int)B, (int)H, (int)L, cur_stream()));
+  return dq;
+}
+
 std::vector<torch::Tensor> masked_pool_fwd(torch::Tensor x,
                                            c10::optional<torch::Tensor> mask) {
   check_bf16(x, "x");
@@ -431,6 +444,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_bwd_fused", &flash_bwd_fused,
         "fused flash bwd: dS + register-accumulated dK/dV");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
+  m.def("flash_dq", &flash_dq, "dQ = dS @ K (MFMA, tr_b16 K^T fragments)");
   m.def("p_from_lse", &p_from_lse, "probabilities from saved logsumexp");
   m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");
   m.def("qkv_repack_bwd3", &qkv_repack_bwd3, "qkv repack bwd from dq,dk,dv");

@@ -41,17 +41,24 @@ def main():
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     print(f"flash_bwd_fused {dt*1e6:.1f} us/call  {4*flops/dt/1e12:.1f} TF-eq")
-    # whole backward chain incl. the one remaining bmm (dq)
+    # whole backward chain incl. the custom dq kernel
     for _ in range(3):
-        dq = torch.matmul(ds, k)
+        dq = ext.flash_dq(ds, k)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        dq = ext.flash_dq(ds, k)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    print(f"flash_dq {dt*1e6:.1f} us/call  {flops/2/dt/1e12:.1f} TF")
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
         ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
-        dq = torch.matmul(ds, k)
+        dq = ext.flash_dq(ds, k)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    print(f"bwd chain (fused + dq bmm) {dt*1e6:.1f} us")
+    print(f"bwd chain (fused + dq) {dt*1e6:.1f} us")
 
 
 if __name__ == "__main__":

@@ -364,3 +364,15 @@ def test_ffn_inference_path_matches_training_path():
     y_train = ffn(x.requires_grad_())
     assert torch.allclose(y_inf.float(), y_train.float(), atol=3e-2,
                           rtol=3e-2), (y_inf.float() - y_train.float()).abs().max()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("B,H,L", [(2, 3, 128), (1, 2, 512), (2, 2, 192)])
+def test_flash_dq_vs_matmul(B, H, L):
+    torch.manual_seed(23)
+    ds = _bf16(torch.randn(B, H, L, L))
+    k = _bf16(torch.randn(B, H, L, 64))
+    dq = ops.hip_ops().flash_dq(ds, k)
+    ref_dq = torch.matmul(ds.float(), k.float())
+    assert torch.allclose(dq.float(), ref_dq, atol=0.5, rtol=3e-2), \
+        (dq.float() - ref_dq).abs().max()

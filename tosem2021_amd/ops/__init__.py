@@ -252,7 +252,7 @@ class _FlashAttentionFn(torch.autograd.Function):
             ddot = hip_ops().fa_dot(do, o)
             ds, dk, dv = hip_ops().flash_bwd_fused(q, k, v, do, mask, lse,
                                                    ddot, scale)
-            dq = torch.matmul(ds, k)  # [q,kv] @ [kv,d]
+            dq = hip_ops().flash_dq(ds, k)  # [q,kv] @ [kv,d], custom MFMA
             return dq, dk, dv, None, None
         s = torch.matmul(q, k.transpose(-1, -2))
         p = reference.p_from_lse(s, mask, lse, scale)
