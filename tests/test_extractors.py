@@ -133,3 +133,30 @@ def test_walker_detection(tmp_path):
     assert "pkg/module.py" not in found
     assert "pkg/engine.cc" not in found
     assert "pkg/ui.ts" not in found
+
+
+def test_gtest_local_macro_expansion(tmp_path):
+    """One-level project-local assertion wrappers (kenlm SLOPPY_CHECK_* style,
+    reference src/DeepSpeech .../lm/model_test.cc:11) count as assertions."""
+    from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+    src = r"""
+#define SLOPPY_CHECK_CLOSE(ref, value, tol) BOOST_CHECK_CLOSE( \
+    static_cast<double>(ref), static_cast<double>(value), tol);
+#define MY_OK(x) EXPECT_TRUE(x)
+
+TEST(Wrapped, UsesLocalMacros) {
+  SLOPPY_CHECK_CLOSE(1.0, compute(), 0.001);
+  MY_OK(flag());
+  EXPECT_EQ(1, one());
+}
+"""
+    p = tmp_path / "wrap_test.cc"
+    p.write_text(src)
+    cases = extract_gtest_file(str(p))
+    assert len(cases) == 1
+    kinds = sorted(a.kind for a in cases[0].assertions)
+    names = sorted(a.call_name for a in cases[0].assertions)
+    assert "approx" in kinds          # SLOPPY_CHECK_CLOSE -> BOOST_CHECK_CLOSE
+    assert "BOOST_CHECK_CLOSE" in names
+    assert "EXPECT_TRUE" in names     # MY_OK
+    assert len(cases[0].assertions) == 3

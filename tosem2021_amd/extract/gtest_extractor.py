@@ -22,6 +22,14 @@ RE_BOOST_CASE = re.compile(
 RE_ASSERT = re.compile(
     r"\b((?:EXPECT|ASSERT)_[A-Z_0-9]+|BOOST_(?:CHECK|REQUIRE|WARN)(?:_[A-Z_0-9]+)?)\s*\(")
 RE_DEATH = re.compile(r"DEATH|THROW", re.I)
+# project-local assertion wrappers, e.g. kenlm's
+#   #define SLOPPY_CHECK_CLOSE(ref, value, tol) BOOST_CHECK_CLOSE(...)
+# (reference src/DeepSpeech/v0.9.3/native_client/kenlm/lm/model_test.cc:11):
+# one-level expansion — a #define in the SAME file whose body contains a
+# known assertion macro makes its invocations count as that assertion.
+RE_LOCAL_MACRO = re.compile(
+    r"^[ \t]*#[ \t]*define[ \t]+([A-Za-z_]\w*)[ \t]*\(",
+    re.M)
 
 
 def _match_brace_block(text: str, open_idx: int) -> int:
@@ -89,6 +97,30 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     except OSError:
         return []
     cases: List[TestCase] = []
+    # one-level local macro table: wrapper name -> underlying assertion macro
+    local_asserts = {}
+    for dm in RE_LOCAL_MACRO.finditer(text):
+        mname = dm.group(1)
+        # macro body: to end of line, following backslash continuations
+        i = dm.end()
+        body_lines = []
+        while True:
+            j = text.find("\n", i)
+            if j < 0:
+                j = len(text)
+            line = text[i:j]
+            body_lines.append(line)
+            if line.rstrip().endswith("\\"):
+                i = j + 1
+            else:
+                break
+        mbody = " ".join(body_lines)
+        am = RE_ASSERT.search(mbody)
+        if am and mname not in ("EXPECT", "ASSERT"):
+            local_asserts[mname] = am.group(1)
+    re_local = (re.compile(r"\b(" + "|".join(map(re.escape, local_asserts))
+                           + r")\s*\(")
+                if local_asserts else None)
     marks = []
     for m in RE_TEST_MACRO.finditer(text):
         if m.group(1) != "INSTANTIATE_TEST_SUITE_P":
@@ -119,6 +151,20 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
             assertions.append(Assertion(
                 kind=kind, call_name=call, source=src[:500], lineno=a_line,
                 exception=exc))
+        if re_local is not None:
+            for am in re_local.finditer(body):
+                wrapper = am.group(1)
+                call = local_asserts[wrapper]
+                src = wrapper + _extract_call(body, am.end() - 1)
+                a_line = lineno + body.count("\n", 0, am.start())
+                kind = "raises" if RE_DEATH.search(call) else "unittest"
+                if "NEAR" in call or "FLOAT_EQ" in call \
+                        or "DOUBLE_EQ" in call or "CLOSE" in call:
+                    kind = "approx"
+                assertions.append(Assertion(
+                    kind=kind, call_name=call, source=src[:500],
+                    lineno=a_line, exception=""))
+            assertions.sort(key=lambda a: a.lineno)
         cases.append(TestCase(
             name=name,
             qualname=f"{suite}.{name}",
