@@ -41,6 +41,21 @@ def _read(path: str) -> Optional[pd.DataFrame]:
     return pd.read_csv(path, encoding="utf-8-sig")
 
 
+# RQ1 derivation notes (round-1 forensics, see docs/ROADMAP.md):
+# The published RQ1_tests.csv rows are NOT the 19 strategy columns — they are
+# a fine-grained oracle-check-type breakdown: strategy-ish columns
+# (status_analysis->status_test, negative_test, logical_condition->
+# logical_statement|logical_expression, Null_pointer, value_range_analysis),
+# Error_Type values (value_error..NotImplementedError) and Approximation_Type
+# values (absolute_relative_tolerence, error_bounding, rounding_tolence).
+# Reconstructing each row's stage distribution from taxonomy_test2.csv shows
+# NO single denominator reproduces the published percentages: the implied
+# denominators cluster ~470-690 for the error-type rows (= the ~560
+# error_handling rows, i.e. per-GROUP normalization) but range 2100-8900 for
+# the strategy rows, and runtime_error (TOTAL 33.57, implied n~188) is
+# inconsistent with the 39 RuntimeError rows in the shipped CSV.  The table
+# appears hand-assembled with mixed denominators; compare_rq1 therefore
+# checks schema + correlation rather than cell equality.
 def compare_rq1(ours_csv: str, ref_csv: str) -> dict:
     ours, ref = _read(ours_csv), _read(ref_csv)
     if ours is None or ref is None:
