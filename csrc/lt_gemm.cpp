@@ -13,6 +13,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hipblaslt/hipblaslt.h>
 
+#include <chrono>
 #include <map>
 #include <mutex>
 #include <tuple>
@@ -189,4 +190,67 @@ bool lt_probe_epilogue(long m, long n, long k, long epi) {
   hipblasLtMatmulDescDestroy(desc);
   return st == HIPBLAS_STATUS_SUCCESS && found > 0 &&
          res.state == HIPBLAS_STATUS_SUCCESS;
+}
+
+// Time the top heuristic algos for a bf16 GEMM config on device — tells us
+// whether the heuristic's first choice leaves wall time on the table
+// (hipBLASLt has no offline tuner in this image; TunableOp crashed).
+// Returns {algo_index, ms} pairs sorted by the heuristic, timed in order.
+std::vector<double> lt_bench_algos(long m, long n, long k, bool ta, bool tb,
+                                   long iters) {
+  auto opts = torch::TensorOptions().dtype(torch::kBFloat16)
+                  .device(torch::kCUDA);
+  auto A = torch::randn({ta ? k : m, ta ? m : k}, opts).contiguous();
+  auto Bm = torch::randn({tb ? n : k, tb ? k : n}, opts).contiguous();
+  auto C = torch::empty({m, n}, opts);  // cm [m, n] == rm [n, m]; scratch
+  hipblasLtMatmulDesc_t desc;
+  TORCH_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F,
+                                        HIP_R_32F) == HIPBLAS_STATUS_SUCCESS);
+  int32_t opA = ta ? HIPBLAS_OP_T : HIPBLAS_OP_N;
+  int32_t opB = tb ? HIPBLAS_OP_T : HIPBLAS_OP_N;
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA,
+                                  sizeof(opA));
+  hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB,
+                                  sizeof(opB));
+  auto la = mk_layout(ta ? k : m, ta ? m : k, ta ? k : m);
+  auto lb = mk_layout(tb ? n : k, tb ? k : n, tb ? n : k);
+  auto lc = mk_layout(m, n, m);
+  hipblasLtMatmulPreference_t pref;
+  hipblasLtMatmulPreferenceCreate(&pref);
+  size_t ws = kWorkspace;
+  hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws));
+  hipblasLtMatmulHeuristicResult_t res[16];
+  int found = 0;
+  auto st = hipblasLtMatmulAlgoGetHeuristic(lt_handle(), desc, la, lb, lc,
+                                            lc, pref, 16, res, &found);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  TORCH_CHECK(st == HIPBLAS_STATUS_SUCCESS && found > 0, "no algos");
+  auto wsbuf = torch::empty({(long)kWorkspace},
+                            opts.dtype(torch::kByte));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  float alpha = 1.f, beta = 0.f;
+  std::vector<double> out;
+  for (int i = 0; i < found; ++i) {
+    if (res[i].state != HIPBLAS_STATUS_SUCCESS) { out.push_back(-1.0); continue; }
+    auto run1 = [&] {
+      return hipblasLtMatmul(lt_handle(), desc, &alpha, A.data_ptr(), la,
+                             Bm.data_ptr(), lb, &beta, C.data_ptr(), lc,
+                             C.data_ptr(), lc, &res[i].algo, wsbuf.data_ptr(),
+                             kWorkspace, stream);
+    };
+    if (run1() != HIPBLAS_STATUS_SUCCESS) { out.push_back(-1.0); continue; }
+    hipStreamSynchronize(stream);
+    auto t0 = std::chrono::steady_clock::now();
+    for (long it = 0; it < iters; ++it) run1();
+    hipStreamSynchronize(stream);
+    auto dt = std::chrono::duration<double, std::milli>(
+                  std::chrono::steady_clock::now() - t0).count() / iters;
+    out.push_back(dt);
+  }
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(lc);
+  hipblasLtMatmulDescDestroy(desc);
+  return out;
 }

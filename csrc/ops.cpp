@@ -56,6 +56,7 @@ hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
 torch::Tensor lt_linear_gelu_bias(torch::Tensor, torch::Tensor,
                                   torch::Tensor);
 bool lt_probe_epilogue(long, long, long, long);
+std::vector<double> lt_bench_algos(long, long, long, bool, bool, long);
 
 namespace {
 
@@ -455,6 +456,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GeLU fwd");
   m.def("lt_linear_gelu_bias", &lt_linear_gelu_bias,
         "hipBLASLt GEMM with fused GELU_BIAS epilogue (inference)");
+  m.def("lt_bench_algos", &lt_bench_algos,
+        "time the heuristic's top algos for a bf16 GEMM config");
   m.def("lt_probe_epilogue", &lt_probe_epilogue,
         "does this hipBLASLt have kernels for (m, n, k, epilogue)?");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GeLU bwd");
