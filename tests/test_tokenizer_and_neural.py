@@ -98,3 +98,13 @@ def test_linear_baseline_tiny(tmp_path):
     # two perfectly-separable synthetic classes: the linear model must learn
     assert res["strategy_micro_f1"] > 0.9, res
     assert res["method_accuracy"] > 0.9
+
+
+def test_mltc_large_config_constructs():
+    import torch
+    from tosem2021_amd.models.classifier import CONFIGS, build_model
+    cfg = CONFIGS["mltc-large"]
+    assert cfg.head_dim == 64  # flash kernel dh
+    m = build_model("mltc-large", dtype=torch.float32)
+    n = sum(p.numel() for p in m.parameters())
+    assert n > 1e9
