@@ -488,6 +488,54 @@ colsum_stage_kernel(const float* __restrict__ ws, float* __restrict__ out,
   *(float4_t*)(out + (long)blockIdx.y * D + col) = s;
 }
 
+// Bias gradient: column-sum of a [N, D] bf16 matrix -> [D] bf16.
+// Two-stage deterministic split-row reduction with f32 accumulation —
+// replaces torch's generic reduce_kernel for the projection bias grads
+// (~58 us -> ~20 us each at [65536, 1024..3072]).
+__global__ void __launch_bounds__(256)
+colsum_bf16_stage1_kernel(const short* __restrict__ x,
+                          float* __restrict__ scratch, long N, int D,
+                          long rows_per_split) {
+  int col = (blockIdx.x * 256 + threadIdx.x) * 8;
+  if (col >= D) return;
+  long r0 = (long)blockIdx.y * rows_per_split;
+  long r1 = min(N, r0 + rows_per_split);
+  float s[8] = {0.f};
+  for (long r = r0; r < r1; ++r) {
+    short8_t v = *(const short8_t*)(x + r * D + col);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] += bf16_to_f32(v[j]);
+  }
+  float* out = scratch + (long)blockIdx.y * D + col;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = s[j];
+}
+
+__global__ void __launch_bounds__(256)
+colsum_bf16_stage2_kernel(const float* __restrict__ scratch,
+                          short* __restrict__ out, int splits, int D) {
+  int col = blockIdx.x * 256 + threadIdx.x;
+  if (col >= D) return;
+  float s = 0.f;
+  for (int r = 0; r < splits; ++r) s += scratch[(long)r * D + col];
+  out[col] = f32_to_bf16(s);
+}
+
+hipError_t colsum_bf16_launch(const void* x, void* scratch, void* out,
+                              long N, int D, hipStream_t stream) {
+  const int splits = 64;
+  long rows_per_split = (N + splits - 1) / splits;
+  dim3 g1((D / 8 + 255) / 256, splits);
+  colsum_bf16_stage1_kernel<<<g1, 256, 0, stream>>>(
+      (const short*)x, (float*)scratch, N, D, rows_per_split);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return e;
+  dim3 g2((D + 255) / 256);
+  colsum_bf16_stage2_kernel<<<g2, 256, 0, stream>>>(
+      (const float*)scratch, (short*)out, splits, D);
+  return hipGetLastError();
+}
+
 hipError_t ln_fwd_launch(const void* x, const void* res, const void* gamma,
                          const void* beta, void* y, void* s_out, void* mean,
                          void* rstd, int N, int D, float eps, int grid,

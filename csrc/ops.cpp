@@ -51,6 +51,8 @@ hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
 hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
                                   void*, int, int, int, hipStream_t);
 hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
+hipError_t colsum_bf16_launch(const void*, void*, void*, long, int,
+                              hipStream_t);
 }
 
 torch::Tensor lt_linear_gelu_bias(torch::Tensor, torch::Tensor,
@@ -383,6 +385,18 @@ std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
   return {ds, dk, dv};
 }
 
+torch::Tensor bias_grad(torch::Tensor dy) {
+  check_bf16(dy, "dy");
+  const int D = (int)dy.size(-1);
+  const long N = dy.numel() / D;
+  TORCH_CHECK(D % 8 == 0, "D % 8 == 0 required");
+  auto scratch = torch::empty({64, D}, dy.options().dtype(torch::kFloat32));
+  auto out = torch::empty({D}, dy.options());
+  CHECK_HIP(colsum_bf16_launch(dy.data_ptr(), scratch.data_ptr(),
+                               out.data_ptr(), N, D, cur_stream()));
+  return out;
+}
+
 torch::Tensor flash_dq(torch::Tensor ds, torch::Tensor k) {
   check_bf16(ds, "ds"); check_bf16(k, "k");
   const long B = k.size(0), H = k.size(1), L = k.size(2);
@@ -446,6 +460,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flash bwd: dS + register-accumulated dK/dV");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
   m.def("flash_dq", &flash_dq, "dQ = dS @ K (MFMA, tr_b16 K^T fragments)");
+  m.def("bias_grad", &bias_grad, "bf16 column-sum for linear bias grads");
   m.def("p_from_lse", &p_from_lse, "probabilities from saved logsumexp");
   m.def("qkv_repack", &qkv_repack, "qkv layout repack (fwd/bwd)");
   m.def("qkv_repack_bwd3", &qkv_repack_bwd3, "qkv repack bwd from dq,dk,dv");

@@ -376,3 +376,39 @@ def test_flash_dq_vs_matmul(B, H, L):
     ref_dq = torch.matmul(ds.float(), k.float())
     assert torch.allclose(dq.float(), ref_dq, atol=0.5, rtol=3e-2), \
         (dq.float() - ref_dq).abs().max()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("N,D", [(4096, 1024), (513, 3072)])
+def test_bias_grad_vs_sum(N, D):
+    torch.manual_seed(24)
+    dy = _bf16(torch.randn(N, D))
+    db = ops.hip_ops().bias_grad(dy)
+    ref_db = dy.float().sum(0)
+    assert torch.allclose(db.float(), ref_db, atol=0.3 + 0.02 * N ** 0.5,
+                          rtol=2e-2), (db.float() - ref_db).abs().max()
+
+
+@pytest.mark.gpu
+def test_fused_linear_autograd_vs_torch():
+    torch.manual_seed(25)
+    N, D, M = 512, 256, 128
+    x = torch.randn(N, D)
+    w = torch.randn(M, D) * 0.05
+    b = torch.randn(M) * 0.1
+    dy = torch.randn(N, M)
+    xg = _bf16(x).requires_grad_()
+    wg = _bf16(w).requires_grad_()
+    bg = _bf16(b).requires_grad_()
+    y = ops.fused_linear(xg, wg, bg)
+    y.backward(_bf16(dy))
+    xr = x.requires_grad_()
+    wr = w.requires_grad_()
+    br = b.requires_grad_()
+    yr = torch.nn.functional.linear(xr, wr, br)
+    yr.backward(dy)
+    assert torch.allclose(y.float().cpu(), yr, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(xg.grad.float().cpu(), xr.grad, atol=5e-2,
+                          rtol=5e-2)
+    assert torch.allclose(wg.grad.float().cpu(), wr.grad, atol=0.5, rtol=5e-2)
+    assert torch.allclose(bg.grad.float().cpu(), br.grad, atol=0.5, rtol=5e-2)

@@ -219,6 +219,37 @@ def out_repack(x):
     return _OutRepackFn.apply(x.contiguous())
 
 
+class _LinearFn(torch.autograd.Function):
+    """F.linear with the bias gradient computed by the two-stage bf16
+    column-sum kernel (csrc/layernorm.hip colsum_bf16).  MEASURED NOTE: not
+    used by the model — routing the projections through this Function cost
+    13.7 ms/step because the explicit dgrad/wgrad matmuls here dispatch to
+    slower hipBLASLt algos than torch's addmm backward (which reaches the
+    tuned split-K Custom_*SK3 kernels); the ~38 us/call bias-grad saving
+    cannot pay for that.  Kept as a standalone op (serving/other shapes)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        return torch.addmm(b, x, w.t())
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.matmul(dy, w)
+        dw = torch.matmul(dy.transpose(0, 1), x)
+        db = hip_ops().bias_grad(dy)
+        return dx, dw, db
+
+
+def fused_linear(x, w, b):
+    """x @ w^T + b with the custom bias-grad path (CUDA bf16 only)."""
+    lead = x.shape[:-1]
+    y = _LinearFn.apply(x.reshape(-1, x.shape[-1]).contiguous(), w, b)
+    return y.view(*lead, -1)
+
+
 def lt_linear_gelu_bias(x, w1, b1):
     """GELU(x @ w1^T + b1) in one hipBLASLt GEMM (GELU_BIAS epilogue) —
     inference only: this hipBLASLt has no aux epilogues (no pre-activation
