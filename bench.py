@@ -71,10 +71,14 @@ def main() -> None:
     from tosem2021_amd.train import TrainConfig, Trainer
 
     cfg = CONFIGS[args.model]
+    if args.seq > cfg.max_seq:
+        # extend the position table for long-sequence runs
+        from tosem2021_amd.models.classifier import MLTCConfig
+        cfg = MLTCConfig(**{**cfg.__dict__, "max_seq": args.seq})
     tcfg = TrainConfig(model=args.model, warmup_steps=0,
                        total_steps=max(args.steps * 100, 1000),
                        bucket_mb=args.bucket_mb)
-    trainer = Trainer(tcfg, device=device)
+    trainer = Trainer(tcfg, device=device, model_cfg=cfg)
 
     tokens, mask, labels = synthetic_batch(
         cfg, args.batch, args.seq, device=device, seed=1234 + rank)

@@ -233,6 +233,12 @@ ln_fwd_t(const short* __restrict__ x, const short* __restrict__ res,
   }
 }
 
+// Deterministic by construction: dgamma/dbeta partials accumulate in
+// registers (each (wave, lane, j) owns a fixed column set) and every wave
+// writes its OWN ws row — no LDS, no atomics, no cross-wave float-order
+// dependence (the earlier LDS-atomicAdd merge was caught non-deterministic
+// by test_kernels_bitwise_deterministic).  ws has gridDim.x * WAVES_PER_BLOCK
+// rows; the two-stage colsum reduces them in fixed order.
 template <int PKTS, bool HASDE>
 __global__ void __launch_bounds__(BLOCK)
 ln_bwd_t(const short* __restrict__ dy, const short* __restrict__ x,
@@ -243,11 +249,6 @@ ln_bwd_t(const short* __restrict__ dy, const short* __restrict__ x,
          int N, int D) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* sg = (float*)smem;
-  float* sb = sg + D;
-  for (int i = threadIdx.x; i < D; i += BLOCK) { sg[i] = 0.f; sb[i] = 0.f; }
-  __syncthreads();
   float gv[PKTS * 8];
 #pragma unroll
   for (int p = 0; p < PKTS; ++p) {
@@ -315,19 +316,18 @@ ln_bwd_t(const short* __restrict__ dy, const short* __restrict__ x,
     if (next >= N) break;
     row = next;
   }
+  float* og = ws_dgamma +
+              (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
+  float* ob = ws_dbeta + (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
 #pragma unroll
   for (int p = 0; p < PKTS; ++p) {
     int base = (p * WAVE + lane) * 8;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sg[base + j], dg_acc[p * 8 + j]);
-      atomicAdd(&sb[base + j], db_acc[p * 8 + j]);
+      og[base + j] = dg_acc[p * 8 + j];
+      ob[base + j] = db_acc[p * 8 + j];
     }
   }
-  __syncthreads();
-  float* og = ws_dgamma + (long)blockIdx.x * D;
-  float* ob = ws_dbeta + (long)blockIdx.x * D;
-  for (int i = threadIdx.x; i < D; i += BLOCK) { og[i] = sg[i]; ob[i] = sb[i]; }
 }
 
 extern "C" {
@@ -575,12 +575,12 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
 #define LNB_T(P)                                                              \
   do {                                                                        \
     if (ds_extra)                                                             \
-      ln_bwd_t<P, true><<<grid, BLOCK, shm, stream>>>(                        \
+      ln_bwd_t<P, true><<<grid, BLOCK, 0, stream>>>(                          \
           (const short*)dy, (const short*)x, (const short*)gamma,             \
           (const float*)mean, (const float*)rstd, (const short*)ds_extra,     \
           (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);             \
     else                                                                      \
-      ln_bwd_t<P, false><<<grid, BLOCK, shm, stream>>>(                       \
+      ln_bwd_t<P, false><<<grid, BLOCK, 0, stream>>>(                         \
           (const short*)dy, (const short*)x, (const short*)gamma,             \
           (const float*)mean, (const float*)rstd, nullptr, (short*)dx,        \
           (float*)ws_dgamma, (float*)ws_dbeta, N, D);                         \

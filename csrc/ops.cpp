@@ -126,12 +126,16 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int D = (int)x.size(-1);
   const long N = x.numel() / D;
   auto dx = torch::empty_like(x);
-  // bwd grid stays modest: the [grid, D] dgamma/dbeta workspace and its
+  // bwd grid stays modest: the [rows, D] dgamma/dbeta workspace and its
   // column-sum scale linearly with the grid
   int grid = (int)std::min<long>((N + 3) / 4, 1024);
+  // the template path (D in {512, 1024}) writes ONE ws row per wave
+  // (deterministic, no atomics); the general path writes one per block
+  const bool templ = (D == 512 || D == 1024);
+  long ws_rows = templ ? (long)grid * 4 : (long)grid;
   auto opts = x.options().dtype(torch::kFloat32);
-  auto ws_dg = torch::empty({grid, D}, opts);
-  auto ws_db = torch::empty({grid, D}, opts);
+  auto ws_dg = torch::empty({ws_rows, D}, opts);
+  auto ws_db = torch::empty({ws_rows, D}, opts);
   const void* de = nullptr;
   if (ds_extra.has_value()) {
     check_bf16(*ds_extra, "ds_extra");
@@ -146,9 +150,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = torch::empty({D}, opts);
   auto scratch = torch::empty({64, D}, opts);
   CHECK_HIP(colsum_launch(ws_dg.data_ptr(), scratch.data_ptr(),
-                          dgamma.data_ptr(), grid, D, cur_stream()));
+                          dgamma.data_ptr(), (int)ws_rows, D, cur_stream()));
   CHECK_HIP(colsum_launch(ws_db.data_ptr(), scratch.data_ptr(),
-                          dbeta.data_ptr(), grid, D, cur_stream()));
+                          dbeta.data_ptr(), (int)ws_rows, D, cur_stream()));
   return {dx, dgamma, dbeta};
 }
 
