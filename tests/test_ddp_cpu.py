@@ -18,7 +18,7 @@ from tosem2021_amd.train import TrainConfig, Trainer
 PORT = 29871
 
 
-def _worker(rank, world, port, out, accum=False):
+def _worker(rank, world, port, out, accum=False, ckpt_dir=None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -26,8 +26,11 @@ def _worker(rank, world, port, out, accum=False):
         torch.manual_seed(100 + rank)  # different init per rank; bcast fixes it
         trainer = Trainer(
             TrainConfig(model="mltc-tiny", lr=1e-3, warmup_steps=0,
-                        dtype="f32", bucket_mb=1),
+                        dtype="f32", bucket_mb=1, ckpt_dir=ckpt_dir),
             device=torch.device("cpu"))
+        if ckpt_dir:
+            trainer.load_or_init()   # crash-resume path: all ranks load the
+                                     # same checkpoint -> consistent state
         cfg = CONFIGS["mltc-tiny"]
         # global batch 8 split across ranks
         per = 8 // world
@@ -48,6 +51,8 @@ def _worker(rank, world, port, out, accum=False):
             else:
                 trainer.step(tokens[sl], mask[sl],
                              {k: v[sl] for k, v in labels.items()})
+        if ckpt_dir and rank == 0:
+            trainer.save()
         out[rank] = trainer.flat.flat.clone()
     finally:
         dist.destroy_process_group()
@@ -122,3 +127,32 @@ def test_ddp_world4_matches_single_process():
         solo.step(tokens, mask, labels)
     diff = (solo.flat.flat - flats[0]).abs().max()
     assert float(diff) < 5e-5, float(diff)
+
+
+@pytest.mark.timeout(300)
+def test_ddp_checkpoint_resume(tmp_path):
+    """Kill-and-resume across a checkpoint (SURVEY.md §5 checkpoint/resume
+    + failure recovery): 2 DDP steps, rank-0 checkpoint, fresh processes
+    resume and take 2 more steps == 4 uninterrupted steps."""
+    world = 2
+    ck = str(tmp_path / "ck")
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        out = mgr.dict()
+        mp.start_processes(_worker, args=(world, PORT + 6, out, False, ck),
+                           nprocs=world, join=True, start_method="spawn")
+        out2 = mgr.dict()
+        mp.start_processes(_worker, args=(world, PORT + 8, out2, False, ck),
+                           nprocs=world, join=True, start_method="spawn")
+        resumed = out2[0].clone()
+        assert torch.equal(out2[0], out2[1])
+
+    torch.manual_seed(100)
+    solo = Trainer(TrainConfig(model="mltc-tiny", lr=1e-3, warmup_steps=0,
+                               dtype="f32"), device=torch.device("cpu"))
+    cfg = CONFIGS["mltc-tiny"]
+    tokens, mask, labels = synthetic_batch(cfg, 8, 32, seed=42)
+    for _ in range(4):
+        solo.step(tokens, mask, labels)
+    diff = (solo.flat.flat - resumed).abs().max()
+    assert float(diff) < 1e-4, float(diff)
