@@ -108,3 +108,38 @@ def test_mltc_large_config_constructs():
     m = build_model("mltc-large", dtype=torch.float32)
     n = sum(p.numel() for p in m.parameters())
     assert n > 1e9
+
+
+def test_tokenizer_invariants():
+    """Property-style invariants: determinism, max_len cap, hash range."""
+    from tosem2021_amd.models.tokenizer import CodeTokenizer, PAD
+    tok = CodeTokenizer(2048)
+    samples = ["assertTrue(x)", "", "  ", "a" * 500,
+               "EXPECT_NEAR(a, b, 1e-5); // unicode \u00e9\u4e2d",
+               "def f():\n    return {1: 'x'}"]
+    for s in samples:
+        ids1 = tok.encode(s, 64)
+        ids2 = tok.encode(s, 64)
+        assert ids1 == ids2                      # deterministic
+        assert len(ids1) <= 64                   # capped
+        assert all(0 <= i < 2048 for i in ids1)  # in vocab range
+    toks, mask = tok.encode_batch(samples, 64)
+    assert toks.shape == mask.shape
+    assert toks.shape[1] <= 64 and toks.shape[1] % 8 == 0
+    assert bool((toks[~mask] == PAD).all())
+
+
+def test_negative_paths():
+    """Corrupt/missing inputs fail loudly, not silently (the corpus'
+    negative_test practice applied to this framework itself)."""
+    import pytest
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    from tosem2021_amd.classify.neural import train_classifier
+    with pytest.raises(Exception):
+        load_taxonomy("/nonexistent/path.csv")
+    with pytest.raises(Exception):
+        train_classifier("/nonexistent/tax.csv", model="mltc-tiny", steps=1,
+                         device="cpu")
+    from tosem2021_amd.models.classifier import build_model
+    with pytest.raises(KeyError):
+        build_model("no-such-config")
