@@ -10,7 +10,7 @@ Regenerates, from any 41-column taxonomy table:
 
 The reference ships only these OUTPUT tables (no analysis code exists in the
 package — SURVEY.md §2); derivations here follow the documented schema and
-are validated in tests/test_analyze_golden.py against the shipped CSVs.
+are validated in tests/test_reference_golden.py against the shipped CSVs.
 """
 from __future__ import annotations
 

@@ -298,10 +298,10 @@ def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
             row.flags = flags
             props = [PROPERTIES[j] for j in range(len(PROPERTIES))
                      if bool(pp[i, j])]
-            if props:
-                row.model = props[0]
-            if len(props) > 1:
-                row.data = props[1]
+            # row_properties reads all four of Model/Data/Code/Oracle — spill
+            # predicted properties across them so nothing is dropped.
+            for slot, val in zip(("model", "data", "code", "oracle"), props):
+                setattr(row, slot, val)
             rows.append(row)
     trainer.model.train()
     return write_taxonomy_csv(rows, out_csv)

@@ -64,7 +64,7 @@ CONFIGS: Dict[str, MLTCConfig] = {
     # small config for fast CPU tests / smoke
     "mltc-tiny": MLTCConfig(vocab_size=512, d_model=128, n_heads=4, n_layers=2,
                             d_ff=256, max_seq=64),
-    # scale-up config (1.1B params): exercises the same kernels at d=2048
+    # scale-up config (1.28B params): exercises the same kernels at d=2048
     # (LN template PKTS=4, 32 flash heads) — sized for 288 GB HBM3E
     "mltc-large": MLTCConfig(d_model=2048, n_heads=32, n_layers=24,
                              d_ff=8192),

@@ -41,7 +41,7 @@ class CodeTokenizer:
     def tokens(self, text: str) -> List[str]:
         out: List[str] = []
         for tok in _TOKEN_RE.findall(text or ""):
-            if self.split_subwords and tok[0].isalpha() or tok[0] == "_":
+            if self.split_subwords and (tok[0].isalpha() or tok[0] == "_"):
                 parts = [p for p in _CAMEL_RE.split(tok) if p]
                 out.extend(p.lower() for p in parts) if len(parts) > 1 \
                     else out.append(tok.lower())

@@ -239,7 +239,7 @@ class _LinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = torch.matmul(dy, w)
         dw = torch.matmul(dy.transpose(0, 1), x)
-        db = hip_ops().bias_grad(dy)
+        db = hip_ops().bias_grad(dy) if dy.is_cuda else dy.sum(0)
         return dx, dw, db
 
 

@@ -54,7 +54,7 @@ class FlatParams:
         self.numel = total
         self.padded = padded
         # f32 optimizer state
-        self.master = self.flat.float()
+        self.master = self.flat.detach().clone().float()
         self.m = torch.zeros(padded, dtype=torch.float32, device=device)
         self.v = torch.zeros(padded, dtype=torch.float32, device=device)
 
