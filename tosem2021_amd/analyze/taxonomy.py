@@ -156,17 +156,21 @@ def row_method(df: pd.DataFrame) -> pd.Series:
     return method
 
 
-def row_stage(df: pd.DataFrame) -> pd.Series:
+def row_stage(df: pd.DataFrame,
+              mapping: Dict[str, str] | None = None) -> pd.Series:
     """Workflow stage per row.
 
     Regenerated taxonomies already carry a stage name in Category; the
     reference master CSV carries open-coding categories that map through
-    CATEGORY_TO_STAGE.
+    CATEGORY_TO_STAGE (default) or an explicit mapping (the RQ1-replication
+    path passes stage_map.RQ1_RECOVERED_CATEGORY_TO_STAGE).
     """
+    m = CATEGORY_TO_STAGE if mapping is None else mapping
+
     def to_stage(c) -> str:
         c = _clean_str(c)
         if c in STAGES:
             return c
-        return CATEGORY_TO_STAGE.get(c, "config_utility")
+        return m.get(c, "config_utility")
 
     return df["Category"].map(to_stage)

@@ -29,13 +29,17 @@ def cmd_mine(args):
 
 
 def cmd_analyze(args):
-    from tosem2021_amd.analyze.tables import write_all
     from tosem2021_amd.analyze.taxonomy import load_taxonomy
     df = load_taxonomy(args.taxonomy)
-    paths = write_all(df, args.out)
-    if not args.no_figures:
-        from tosem2021_amd.analyze.figures import write_figures
-        paths.update(write_figures(df, args.out))
+    if args.legacy_tables:
+        from tosem2021_amd.analyze.tables import write_all
+        paths = write_all(df, args.out)
+        if not args.no_figures:
+            from tosem2021_amd.analyze.figures import write_figures
+            paths.update(write_figures(df, args.out))
+    else:
+        from tosem2021_amd.analyze.mirror import write_mirror
+        paths = write_mirror(df, args.out)
     for k, p in paths.items():
         print(f"{k}: {p}")
 
@@ -53,8 +57,12 @@ def cmd_agreement(args):
 
 
 def cmd_golden(args):
-    from tosem2021_amd.analyze.golden import golden_diff
-    res = golden_diff(args.ours, args.reference)
+    if args.mirror:
+        from tosem2021_amd.analyze.golden_mirror import mirror_diff
+        res = mirror_diff(args.ours, args.reference)
+    else:
+        from tosem2021_amd.analyze.golden import golden_diff
+        res = golden_diff(args.ours, args.reference)
     print(json.dumps(res, indent=2))
     sys.exit(0 if res["ok"] else 1)
 
@@ -118,10 +126,14 @@ def main(argv=None):
     p.add_argument("--out", required=True)
     p.set_defaults(fn=cmd_mine)
 
-    p = sub.add_parser("analyze", help="regenerate RQ1/RQ3/RQ4 tables+figures")
+    p = sub.add_parser("analyze", help="regenerate the full RQs/ artifact "
+                       "mirror (every shipped CSV + SVG plots)")
     p.add_argument("--taxonomy", required=True)
     p.add_argument("--out", required=True)
     p.add_argument("--no-figures", action="store_true")
+    p.add_argument("--legacy-tables", action="store_true",
+                   help="emit the round-1 analysis tables instead of the "
+                        "file-for-file mirror")
     p.set_defaults(fn=cmd_analyze)
 
     p = sub.add_parser("agreement", help="score rules vs the reference labels")
@@ -133,6 +145,8 @@ def main(argv=None):
     p = sub.add_parser("golden", help="diff regenerated tables vs reference")
     p.add_argument("--ours", required=True)
     p.add_argument("--reference", required=True)
+    p.add_argument("--mirror", action="store_true",
+                   help="compare the complete file-for-file RQs/ mirror")
     p.set_defaults(fn=cmd_golden)
 
     p = sub.add_parser("report", help="print a taxonomy summary report")
