@@ -1,0 +1,120 @@
+"""Tests for the stdlib XLSX reader (utils/xlsx.py) and the L2 labeling
+ingestion (corpus/labels.py) — VERDICT round-1 missing item 3."""
+import os
+import zipfile
+
+import pytest
+
+
+def _make_xlsx(path, rows, shared=("alpha", "beta")):
+    """Hand-build a minimal xlsx: shared strings + one sheet."""
+    ss = ('<?xml version="1.0"?><sst xmlns="http://schemas.openxmlformats.'
+          'org/spreadsheetml/2006/main" count="2" uniqueCount="2">'
+          + "".join(f"<si><t>{s}</t></si>" for s in shared) + "</sst>")
+    sheet_rows = []
+    for i, row in enumerate(rows, start=1):
+        cells = []
+        for j, (ctype, val) in enumerate(row):
+            ref = chr(ord("A") + j) + str(i)
+            if ctype == "s":
+                cells.append(f'<c r="{ref}" t="s"><v>{val}</v></c>')
+            elif ctype == "inline":
+                cells.append(f'<c r="{ref}" t="inlineStr"><is><t>{val}</t>'
+                             f'</is></c>')
+            elif ctype == "skip":
+                continue
+            else:
+                cells.append(f'<c r="{ref}"><v>{val}</v></c>')
+        sheet_rows.append(f'<row r="{i}">' + "".join(cells) + "</row>")
+    sheet = ('<?xml version="1.0"?><worksheet xmlns="http://schemas.'
+             'openxmlformats.org/spreadsheetml/2006/main"><sheetData>'
+             + "".join(sheet_rows) + "</sheetData></worksheet>")
+    wb = ('<?xml version="1.0"?><workbook xmlns="http://schemas.'
+          'openxmlformats.org/spreadsheetml/2006/main" xmlns:r="http://'
+          'schemas.openxmlformats.org/officeDocument/2006/relationships">'
+          '<sheets><sheet name="S1" sheetId="1" r:id="rId1"/></sheets>'
+          '</workbook>')
+    rels = ('<?xml version="1.0"?><Relationships xmlns="http://schemas.'
+            'openxmlformats.org/package/2006/relationships">'
+            '<Relationship Id="rId1" Type="http://schemas.openxmlformats.'
+            'org/officeDocument/2006/relationships/worksheet" '
+            'Target="worksheets/sheet1.xml"/></Relationships>')
+    with zipfile.ZipFile(path, "w") as zf:
+        zf.writestr("xl/workbook.xml", wb)
+        zf.writestr("xl/_rels/workbook.xml.rels", rels)
+        zf.writestr("xl/sharedStrings.xml", ss)
+        zf.writestr("xl/worksheets/sheet1.xml", sheet)
+
+
+def test_xlsx_reader_roundtrip(tmp_path):
+    from tosem2021_amd.utils.xlsx import first_sheet, read_xlsx
+    p = str(tmp_path / "t.xlsx")
+    _make_xlsx(p, [
+        [("s", 0), ("n", 42), ("inline", "hello")],
+        [("skip", None), ("s", 1)],          # sparse row: A2 missing
+    ])
+    sheets = read_xlsx(p)
+    assert list(sheets) == ["S1"]
+    rows = sheets["S1"]
+    assert rows[0] == ["alpha", "42", "hello"]
+    assert rows[1] == ["", "beta", ""]       # gap filled, width padded
+    assert first_sheet(p) == rows
+
+
+def test_xlsx_reader_sheet_filter(tmp_path):
+    from tosem2021_amd.utils.xlsx import read_xlsx
+    p = str(tmp_path / "t.xlsx")
+    _make_xlsx(p, [[("n", 1)]])
+    assert read_xlsx(p, sheet="nope") == {}
+
+
+@pytest.fixture(scope="module")
+def reference_mounted():
+    if not os.path.isdir("/root/reference/selection/completed-labels"):
+        pytest.skip("reference corpus not mounted on this box")
+
+
+def test_release_sheets_load(reference_mounted):
+    from tosem2021_amd.corpus.labels import load_all_release_sheets
+    sheets = load_all_release_sheets()
+    # all 8 labeling artifacts ingested (SURVEY.md §2.2 manual labels row)
+    assert set(sheets) == {"ray", "apollo", "DeepSpeech", "auto-sklearn",
+                           "autokeras", "automl", "nupic", "tpot"}
+    assert len(sheets["apollo"]) > 800            # 884 tracked files
+    assert len(sheets["auto-sklearn"]) == 123
+    ask = sheets["auto-sklearn"]
+    labeled = [r for r in ask if r.test_type]
+    assert len(labeled) > 50                      # open codes present
+    assert any("error" in r.test_type.lower() for r in labeled)
+
+
+def test_codebook_covers_strategy_vocabulary(reference_mounted):
+    from tosem2021_amd.corpus.labels import (
+        codebook_strategy_coverage, load_codebook)
+    cb = load_codebook()
+    assert len(cb) > 500                           # 590 instrument tags
+    cov = codebook_strategy_coverage(cb)
+    # every strategy except the status_analysis family is grounded in the
+    # instrument, and no codebook category is unmapped
+    assert cov["strategies_uncovered"] == []
+    assert cov["unknown_categories"] == []
+
+
+def test_file_id_lineage_l2_to_l3(reference_mounted):
+    """Every FileID in the master taxonomy traces to its project's
+    labeling sheet (100% coverage measured round 2)."""
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    from tosem2021_amd.corpus.labels import (
+        lineage_check, load_all_release_sheets)
+    df = load_taxonomy("/root/reference/RQs/taxonomy_test2.csv")
+    res = lineage_check(df, load_all_release_sheets())
+    assert len(res) == 8
+    for repo, r in res.items():
+        assert r["coverage"] == 1.0, (repo, r)
+
+
+def test_case_labels_sheet(reference_mounted):
+    from tosem2021_amd.corpus.labels import load_case_labels
+    cases = load_case_labels()
+    assert len(cases) > 2000
+    assert any("error-handling" in c["label"].lower() for c in cases)

@@ -74,3 +74,28 @@ def test_funnel_cli(reference_root, tmp_path):
     assert r.returncode == 0, r.stderr[-1000:]
     assert "input: 311 candidates" in r.stdout
     assert os.path.exists(tmp_path / "round_2.csv")
+
+
+def test_funnel_forensics_golden():
+    """Round-2 forensics of the shipped selection tables (selection.py
+    module docstring): the funnels are two disjoint streams, not nested
+    refinements; the recovered metric screen captures every in-table
+    finalist."""
+    import os
+
+    import pytest
+
+    reposition = "/root/reference/selection/Reposition"
+    if not os.path.isdir(reposition):
+        pytest.skip("reference corpus not mounted on this box")
+    from tosem2021_amd.corpus.selection import funnel_forensics
+    res = funnel_forensics(reposition)
+    assert res["v2_rows"] == 225 and res["v2_unique"] == 157
+    assert res["v2_subset_of_v3"]                # v2 = per-topic precursor
+    assert res["v3_rows"] == 311
+    assert res["v4_rows"] == 28 and res["v4_overlap_v3"] == 0  # disjoint
+    assert res["finalists"] == 14 and res["finalists_in_v3"] == 13
+    assert res["finalist_outside_v3"] == ["MycroftAI/mycroft-core"]
+    # the recovered screen: 28 candidates, all 13 in-table finalists kept
+    assert res["screen_selects"] == 28
+    assert res["screen_captures_finalists"] == 13

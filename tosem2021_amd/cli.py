@@ -67,6 +67,33 @@ def cmd_golden(args):
     sys.exit(0 if res["ok"] else 1)
 
 
+def cmd_labels(args):
+    """Ingest the L2 manual-labeling artifacts (XLSX/CSV sheets + codebooks)
+    and report lineage/coverage against the master taxonomy."""
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    from tosem2021_amd.corpus.labels import (
+        codebook_strategy_coverage, lineage_check, load_all_release_sheets,
+        load_case_labels, load_codebook)
+    sheets = load_all_release_sheets(args.reference)
+    cb = load_codebook(args.reference)
+    res = {
+        "release_sheets": {
+            k: {"rows": len(v),
+                "with_file_id": sum(1 for r in v if r.file_id is not None),
+                "with_test_type": sum(1 for r in v if r.test_type)}
+            for k, v in sheets.items()},
+        "codebook": codebook_strategy_coverage(cb),
+        "case_labels": len(load_case_labels(args.reference)),
+    }
+    if args.taxonomy:
+        df = load_taxonomy(args.taxonomy)
+        res["lineage"] = lineage_check(df, sheets)
+    print(json.dumps(res, indent=2))
+    if args.json:
+        with open(args.json, "w") as f:
+            json.dump(res, f, indent=2)
+
+
 def cmd_report(args):
     from tosem2021_amd.analyze.report import summary_report
     from tosem2021_amd.analyze.taxonomy import load_taxonomy
@@ -148,6 +175,14 @@ def main(argv=None):
     p.add_argument("--mirror", action="store_true",
                    help="compare the complete file-for-file RQs/ mirror")
     p.set_defaults(fn=cmd_golden)
+
+    p = sub.add_parser("labels", help="ingest the L2 labeling sheets + "
+                       "codebooks; report L2->L3 lineage")
+    p.add_argument("--reference", default="/root/reference")
+    p.add_argument("--taxonomy", default=None,
+                   help="master taxonomy CSV for the lineage check")
+    p.add_argument("--json", default=None)
+    p.set_defaults(fn=cmd_labels)
 
     p = sub.add_parser("report", help="print a taxonomy summary report")
     p.add_argument("--taxonomy", required=True)
