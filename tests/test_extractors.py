@@ -160,3 +160,88 @@ TEST(Wrapped, UsesLocalMacros) {
     assert "BOOST_CHECK_CLOSE" in names
     assert "EXPECT_TRUE" in names     # MY_OK
     assert len(cases[0].assertions) == 3
+
+
+def test_ts_nested_describe_and_body_extent(tmp_path):
+    """Round-2 hardening: brace-matched it() bodies (no assertion bleed
+    between cases) and full nested-describe suite paths."""
+    from tosem2021_amd.extract.ts_extractor import extract_ts_file
+    src = """
+describe('outer', () => {
+    describe('inner', () => {
+        it('first', () => {
+            expect(a).to.equal(1);
+        });
+        // helper between cases — must not attach to 'first'
+        const check = () => { assert.ok(stray()); };
+        it('second', async () => {
+            expect(b).to.equal(2);
+            expect(c).to.throw();
+        });
+    });
+    it('outer-level', () => {
+        assert.strictEqual(d, 4);
+    });
+});
+"""
+    p = tmp_path / "x.test.ts"
+    p.write_text(src)
+    cases = extract_ts_file(str(p))
+    byname = {c.name: c for c in cases}
+    assert set(byname) == {"first", "second", "outer-level"}
+    assert byname["first"].qualname == "outer.inner.first"
+    assert byname["second"].qualname == "outer.inner.second"
+    assert byname["outer-level"].qualname == "outer.outer-level"
+    # 'first' must hold exactly its own assertion; the stray helper assert
+    # between the cases belongs to neither
+    assert len(byname["first"].assertions) == 1
+    assert len(byname["second"].assertions) == 2
+    assert len(byname["outer-level"].assertions) == 1
+    assert "stray" not in byname["first"].source
+
+
+def test_ts_braceless_arrow_case(tmp_path):
+    from tosem2021_amd.extract.ts_extractor import extract_ts_file
+    src = """
+it('compact', () => expect(x).to.equal(1));
+it('next', () => { expect(y).to.equal(2); });
+"""
+    p = tmp_path / "y.test.ts"
+    p.write_text(src)
+    cases = extract_ts_file(str(p))
+    assert len(cases) == 2
+    assert len(cases[0].assertions) == 1
+    assert len(cases[1].assertions) == 1
+
+
+def test_gtest_instantiate_multiplicity(tmp_path):
+    """TEST_P cases carry the INSTANTIATE_* count as param multiplicity
+    (each instantiation runs the whole suite once)."""
+    from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+    src = """
+class ParamSuite : public ::testing::TestWithParam<int> {};
+
+TEST_P(ParamSuite, Works) {
+  EXPECT_GT(GetParam(), 0);
+}
+TEST_P(ParamSuite, AlsoWorks) {
+  EXPECT_LT(GetParam(), 100);
+}
+TEST(PlainSuite, One) { EXPECT_TRUE(ok()); }
+
+INSTANTIATE_TEST_SUITE_P(Small, ParamSuite, ::testing::Values(1, 2));
+INSTANTIATE_TEST_CASE_P(Legacy, ParamSuite, ::testing::Values(3));
+"""
+    p = tmp_path / "p_test.cc"
+    p.write_text(src)
+    cases = {c.name: c for c in extract_gtest_file(str(p))}
+    assert set(cases) == {"Works", "AlsoWorks", "One"}
+    assert cases["Works"].is_parametrized
+    assert cases["Works"].param_multiplicity == 2
+    assert cases["AlsoWorks"].param_multiplicity == 2
+    assert not cases["One"].is_parametrized
+    assert cases["One"].param_multiplicity == 1
+    # classify_case folds the multiplicity into the Cases column
+    from tosem2021_amd.classify.rules import classify_case
+    rows = classify_case(cases["Works"], repo="Apollo", file_id=1)
+    assert rows[0].cases == 2     # 1 assertion x 2 instantiations

@@ -278,7 +278,8 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
     head.repo = repo
     head.file_id = file_id
     head.component = component or case.file_rel
-    head.cases = max(len(case.assertions), 1)
+    head.cases = max(len(case.assertions), 1) \
+        * max(case.param_multiplicity, 1)
     if case.uses_mock:
         head.flags["mock_test"] = 1
     rows.append(head)

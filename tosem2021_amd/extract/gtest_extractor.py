@@ -13,8 +13,14 @@ from typing import List, Optional
 from tosem2021_amd.extract.python_extractor import Assertion, TestCase
 
 RE_TEST_MACRO = re.compile(
-    r"\b(TEST|TEST_F|TEST_P|TYPED_TEST|TYPED_TEST_P|INSTANTIATE_TEST_SUITE_P)"
+    r"\b(TEST|TEST_F|TEST_P|TYPED_TEST|TYPED_TEST_P)"
     r"\s*\(\s*([A-Za-z_]\w*)\s*,\s*([A-Za-z_]\w*)\s*\)")
+# INSTANTIATE_TEST_SUITE_P(Prefix, Suite, ...) and the pre-1.10 _CASE_P
+# spelling, plus the typed variants: each one names a suite whose TEST_P /
+# TYPED_TEST_P cases it multiplies (arg 2 = the suite).
+RE_INSTANTIATE = re.compile(
+    r"\bINSTANTIATE_(?:TYPED_)?TEST_(?:SUITE|CASE)_P"
+    r"\s*\(\s*([A-Za-z_]\w*)\s*,\s*([A-Za-z_]\w*)")
 # boost.test (kenlm, openfst in the DeepSpeech snapshot): one-arg case macro
 RE_BOOST_CASE = re.compile(
     r"\b(BOOST_AUTO_TEST_CASE|BOOST_FIXTURE_TEST_CASE|BOOST_AUTO_TEST_CASE_TEMPLATE)"
@@ -121,10 +127,14 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     re_local = (re.compile(r"\b(" + "|".join(map(re.escape, local_asserts))
                            + r")\s*\(")
                 if local_asserts else None)
+    # per-suite instantiation counts (TEST_P multiplicity, VERDICT r1
+    # missing item 6): each INSTANTIATE_* runs the whole suite once more
+    inst_counts: dict = {}
+    for im in RE_INSTANTIATE.finditer(text):
+        inst_counts[im.group(2)] = inst_counts.get(im.group(2), 0) + 1
     marks = []
     for m in RE_TEST_MACRO.finditer(text):
-        if m.group(1) != "INSTANTIATE_TEST_SUITE_P":
-            marks.append((m, m.group(2), m.group(3)))
+        marks.append((m, m.group(2), m.group(3)))
     for m in RE_BOOST_CASE.finditer(text):
         marks.append((m, "", m.group(2)))
     for m, suite, name in sorted(marks, key=lambda t: t[0].start()):
@@ -165,6 +175,8 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
                     kind=kind, call_name=call, source=src[:500],
                     lineno=a_line, exception=""))
             assertions.sort(key=lambda a: a.lineno)
+        macro = m.group(1) if m.re is RE_TEST_MACRO else ""
+        parametrized = macro in ("TEST_P", "TYPED_TEST_P")
         cases.append(TestCase(
             name=name,
             qualname=f"{suite}.{name}",
@@ -175,5 +187,8 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
             assertions=assertions,
             uses_mock=bool(re.search(r"\bMOCK_METHOD|NiceMock|StrictMock|gmock",
                                      body)),
+            is_parametrized=parametrized,
+            param_multiplicity=max(inst_counts.get(suite, 0), 1)
+            if parametrized else 1,
         ))
     return cases

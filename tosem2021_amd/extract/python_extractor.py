@@ -36,6 +36,9 @@ class TestCase:
     markers: List[str] = field(default_factory=list)    # pytest marks
     uses_mock: bool = False
     is_parametrized: bool = False
+    # gtest TEST_P/TYPED_TEST_P: number of INSTANTIATE_* macros naming this
+    # suite (each runs the whole suite once); 1 for plain cases
+    param_multiplicity: int = 1
 
 
 UNITTEST_ASSERT_PREFIX = "assert"
