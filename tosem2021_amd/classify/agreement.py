@@ -77,16 +77,24 @@ def evaluate_rules_on_taxonomy(df: pd.DataFrame, limit: int = 0) -> dict:
     pred_method: List[str] = []
     texts = df["Labels"].astype(str).tolist()
     comps = df["Component"].astype(str).tolist()
-    for text, comp in zip(texts, comps):
+    repos = df["Repo"].astype(str).tolist()
+    from tosem2021_amd.classify.property_lexicon import (
+        default_lexicon, property_features)
+    lex = default_lexicon()
+    for text, comp, repo in zip(texts, comps, repos):
         row = classify_text(text, name="", path=comp)
         pred_strat.append(set(row.strategies()))
-        pred_props.append(set(row.properties()))
+        if lex is not None:
+            pred_props.append(set(lex.predict(
+                property_features(text, comp, repo, row=row))))
+        else:
+            pred_props.append(set(row.properties()))
         pred_method.append(row.method)
 
     strat_scores = _score_sets(pred_strat, gold_strat, STRATEGIES)
     prop_scores = _score_sets(pred_props, gold_props, PROPERTIES)
     method_acc = sum(p == g for p, g in zip(pred_method, gold_method)) / len(df)
-    return {
+    res = {
         "n_rows": len(df),
         "strategy": {l: {"precision": s.precision, "recall": s.recall,
                          "f1": s.f1, "support": s.tp + s.fn}
@@ -94,7 +102,17 @@ def evaluate_rules_on_taxonomy(df: pd.DataFrame, limit: int = 0) -> dict:
         "strategy_micro_f1": micro_f1(strat_scores),
         "property_micro_f1": micro_f1(prop_scores),
         "method_accuracy": method_acc,
+        "property_labeler": "lexicon" if lex is not None else "regex",
     }
+    if lex is not None:
+        # the committed lexicon was fit on the even-index gold rows
+        # (property_lexicon.py protocol) — report the uncontaminated
+        # held-out (odd-row) property score alongside the full-set one
+        odd = [i for i in range(len(df)) if i % 2 == 1]
+        prop_ho = _score_sets([pred_props[i] for i in odd],
+                              [gold_props[i] for i in odd], PROPERTIES)
+        res["property_micro_f1_heldout"] = micro_f1(prop_ho)
+    return res
 
 
 def report(result: dict) -> str:

@@ -275,6 +275,7 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
     head = classify_text(
         f"{desc}: " + "; ".join(a.source for a in case.assertions[:4]),
         name=case.qualname, path=case.file_rel)
+    from tosem2021_amd.classify.property_lexicon import apply_to_row
     head.repo = repo
     head.file_id = file_id
     head.component = component or case.file_rel
@@ -282,12 +283,14 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
         * max(case.param_multiplicity, 1)
     if case.uses_mock:
         head.flags["mock_test"] = 1
+    apply_to_row(head, desc, head.component, repo)
     rows.append(head)
     for a in case.assertions:
         r = classify_text(a.source, name=case.qualname, path=case.file_rel)
         r.repo = repo
         r.file_id = file_id
         r.component = component or case.file_rel
+        apply_to_row(r, a.source, r.component, repo)
         # assertion rows inherit the enclosing case's workflow stage (an
         # assertion's own text rarely carries stage cues)
         r.category = head.category
