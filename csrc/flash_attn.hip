@@ -426,8 +426,10 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
       pv[r] = __expf(s_acc[r] * scale + mb - lse_lds[ql]);
       gv[r] = scale * pv[r] * (dp_acc[r] - dd_lds[ql]);
     }
-    // dS out, natural [q, kv] rows (2 B per lane, lanes contiguous in kv)
-    if (kv_valid) {
+    // dS out, natural [q, kv] rows (2 B per lane, lanes contiguous in kv);
+    // skipped entirely when ds == nullptr (round 2: dQ is recomputed by
+    // flash_dq_recompute_kernel, no dS materialization)
+    if (ds != nullptr && kv_valid) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int ql = (r & 3) + 8 * (r >> 2) + 4 * half;
@@ -706,6 +708,249 @@ extern "C" hipError_t flash_dq_launch(const void* ds, const void* k, void* dq,
   dim3 grid(B * H * n_qblocks);
   flash_dq_kernel<<<grid, FA_BLOCK, K_LDS_BYTES, stream>>>(
       (const short*)ds, (const short*)k, (short*)dq, B, H, L);
+  return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
+// dQ by recompute (round 2): forward-orientation kernel that recomputes
+// S = QK^T and dP = dO V^T per kv tile, forms dS in registers and
+// accumulates dQ = dS @ K — no dS materialization at all.  Replaces the
+// bwd pipeline's [B, H, L, L] bf16 dS HBM round-trip (1.07 GB written by
+// flash_bwd_fused + 1.07 GB re-read by flash_dq at the bench shape) and
+// the separate flash_dq pass; flash_bwd_fused now only emits dK/dV.
+//
+// Structure mirrors flash_fwd exactly (two independent 32-row q-blocks per
+// wave for ILP; K/V tiles staged swizzled in LDS, double-staged across the
+// kv loop):
+//   S  C-layout (reg=kv, lane=q) from mfma(A=K-frag, B=Q-rows)
+//   dP C-layout (reg=kv, lane=q) from mfma(A=V-frag, B=dO-rows)
+//   lse/ddot are LANE-local scalars (one q row per lane), the mask bias
+//   indexes by the kv REGISTER — both orientations line up for free
+//   dS = scale * P * (dP - D) in registers
+//   dS -> A-fragments [row=q][k=kv] via the same cvt_pk_bf16 +
+//   permlane32_swap repack the fwd uses for P
+//   dQ[q, d] += mfma(dS-frag, K^T-frag), K^T gathered ds_read_b64_tr_b16
+//   from the staged K tile (same law as the fwd V^T gather).
+extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
+flash_dq_recompute_kernel(const short* __restrict__ q,
+                          const short* __restrict__ k,
+                          const short* __restrict__ v,
+                          const short* __restrict__ dout,
+                          const float* __restrict__ mask,
+                          const float* __restrict__ lse,
+                          const float* __restrict__ ddot,
+                          short* __restrict__ dq,
+                          int B, int H, int L, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+
+  const int rows_per_wg = 2 * FA_QWG;                 // 256 q rows
+  const int n_qblocks = (L + rows_per_wg - 1) / rows_per_wg;
+  int bid = xcd_group_remap(blockIdx.x, gridDim.x, n_qblocks);
+  int bh = bid / n_qblocks;
+  int qb = bid % n_qblocks;
+  const int b = bh / H;
+  const long bh_off = (long)bh * L * FA_DH;
+  const int q_baseA = qb * rows_per_wg + wid * FA_QB;
+  const int q_baseB = q_baseA + FA_QWG;
+  const int my_qA = q_baseA + col;
+  const int my_qB = q_baseB + col;
+  const bool validA = my_qA < L;
+  const bool validB = my_qB < L;
+  const float* mrow = mask ? mask + (long)b * L : nullptr;
+
+  short8_t qfA[4], qfB[4], dofA[4], dofB[4];
+  {
+    const short* qrA = q + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
+    const short* qrB = q + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
+    const short* drA = dout + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
+    const short* drB = dout + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      qfA[c] = *(const short8_t*)(qrA + c * 16 + half * 8);
+      qfB[c] = *(const short8_t*)(qrB + c * 16 + half * 8);
+      dofA[c] = *(const short8_t*)(drA + c * 16 + half * 8);
+      dofB[c] = *(const short8_t*)(drB + c * 16 + half * 8);
+    }
+  }
+  const float lseA = lse[(long)bh * L + (validA ? my_qA : L - 1)];
+  const float lseB = lse[(long)bh * L + (validB ? my_qB : L - 1)];
+  const float ddA = ddot[(long)bh * L + (validA ? my_qA : L - 1)];
+  const float ddB = ddot[(long)bh * L + (validB ? my_qB : L - 1)];
+
+  f32x16 dqA[2], dqB[2];
+#pragma unroll
+  for (int t = 0; t < 2; ++t) { dqA[t] = (f32x16)(0.f); dqB[t] = (f32x16)(0.f); }
+
+  const int n_kv = L / FA_KVB;
+  const int srow = tid >> 3, sc8 = (tid & 7) * 16;
+  short8_t kv8 = *(const short8_t*)(k + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
+  short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
+                                    (sc8 >> 1));
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int kv0 = kt * FA_KVB;
+    __syncthreads();
+    {
+      *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
+      *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
+    }
+    __syncthreads();
+    if (kt + 1 < n_kv) {
+      kv8 = *(const short8_t*)(k + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + bh_off +
+                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+    }
+
+    // ---- S and dP for BOTH q-blocks (16 back-to-back MFMAs) ----
+    f32x16 sA = (f32x16)(0.f), sB = (f32x16)(0.f);
+    f32x16 pA = (f32x16)(0.f), pB = (f32x16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int byte_off = (16 * c + 8 * half) * 2;
+      short8_t kf = *(const short8_t*)((char*)k_lds + col * 128 +
+                                       kswz(col, byte_off));
+      short8_t vf = *(const short8_t*)((char*)v_lds + col * 128 +
+                                       kswz(col, byte_off));
+      sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfA[c], sA, 0, 0, 0);
+      sB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfB[c], sB, 0, 0, 0);
+      pA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofA[c], pA, 0, 0, 0);
+      pB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofB[c], pB, 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- dS in registers (lse/ddot lane-local, mask by kv register) ----
+    float gA[16], gB[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+      float mb = mrow ? mrow[kv0 + kv_local] : 0.f;
+      float expA = __expf(sA[r] * scale + mb - lseA);
+      float expB = __expf(sB[r] * scale + mb - lseB);
+      gA[r] = scale * expA * (pA[r] - ddA);
+      gB[r] = scale * expB * (pB[r] - ddB);
+    }
+
+    // ---- dS -> A-fragments [row=q][k=kv] (fwd P-repack idiom) ----
+    short8_t gfA[2], gfB[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+      uint4_t uA, uB;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int r0 = c * 8 + 2 * i;
+        int r1 = c * 8 + 4 + 2 * i;
+        unsigned loA, hiA, loB, hiB;
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(loA)
+            : "v"(gA[r0]), "v"(gA[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hiA)
+            : "v"(gA[r1]), "v"(gA[r1 + 1]));
+        auto swA = __builtin_amdgcn_permlane32_swap(loA, hiA, false, false);
+        uA[i] = swA[0]; uA[i + 2] = swA[1];
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(loB)
+            : "v"(gB[r0]), "v"(gB[r0 + 1]));
+        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(hiB)
+            : "v"(gB[r1]), "v"(gB[r1 + 1]));
+        auto swB = __builtin_amdgcn_permlane32_swap(loB, hiB, false, false);
+        uB[i] = swB[0]; uB[i + 2] = swB[1];
+      }
+      gfA[c] = __builtin_bit_cast(short8_t, uA);
+      gfB[c] = __builtin_bit_cast(short8_t, uB);
+    }
+
+    // ---- dQ += dS @ K via tr_b16-gathered K^T fragments ----
+    {
+      typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
+      const unsigned kbase = (unsigned)(unsigned long)(char*)k_lds;
+      const int kv_mate = (lane >> 2) & 3;
+      const int d_lane = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+      unsigned a[8];
+#pragma unroll
+      for (int c = 0; c < 2; ++c)
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
+#pragma unroll
+          for (int t = 0; t < 2; ++t) {
+            int kv = 16 * c + 8 * half + 4 * rr + kv_mate;
+            int dcol = (32 * t + d_lane) * 2;
+            a[c * 4 + rr * 2 + t] =
+                kbase + kv * 128 + (dcol ^ ((kv & 7) << 4));
+          }
+      uint2_t r[8];
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8\n\t"
+          "ds_read_b64_tr_b16 %1, %9\n\t"
+          "ds_read_b64_tr_b16 %2, %10\n\t"
+          "ds_read_b64_tr_b16 %3, %11\n\t"
+          "ds_read_b64_tr_b16 %4, %12\n\t"
+          "ds_read_b64_tr_b16 %5, %13\n\t"
+          "ds_read_b64_tr_b16 %6, %14\n\t"
+          "ds_read_b64_tr_b16 %7, %15\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(r[0]), "=&v"(r[1]), "=&v"(r[2]), "=&v"(r[3]),
+            "=&v"(r[4]), "=&v"(r[5]), "=&v"(r[6]), "=&v"(r[7])
+          : "v"(a[0]), "v"(a[1]), "v"(a[2]), "v"(a[3]), "v"(a[4]), "v"(a[5]),
+            "v"(a[6]), "v"(a[7])
+          : "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          typedef __attribute__((ext_vector_type(4))) unsigned uint4_t;
+          uint4_t w;
+          w[0] = r[c * 4 + 0 * 2 + t][0];
+          w[1] = r[c * 4 + 0 * 2 + t][1];
+          w[2] = r[c * 4 + 1 * 2 + t][0];
+          w[3] = r[c * 4 + 1 * 2 + t][1];
+          short8_t kf = __builtin_bit_cast(short8_t, w);
+          dqA[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(gfA[c], kf,
+                                                           dqA[t], 0, 0, 0);
+          dqB[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(gfB[c], kf,
+                                                           dqB[t], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+  // ---- epilogue ----
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
+      int qA = q_baseA + rloc, qB = q_baseB + rloc;
+      if (qA < L)
+        dq[bh_off + (long)qA * FA_DH + 32 * t + col] = f32_to_bf16(dqA[t][r]);
+      if (qB < L)
+        dq[bh_off + (long)qB * FA_DH + 32 * t + col] = f32_to_bf16(dqB[t][r]);
+    }
+  }
+}
+
+extern "C" hipError_t flash_dq_recompute_launch(
+    const void* q, const void* k, const void* v, const void* dout,
+    const void* mask, const void* lse, const void* ddot, void* dq,
+    int B, int H, int L, float scale, hipStream_t stream) {
+  int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
+  dim3 grid(B * H * n_qblocks);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES;
+  flash_dq_recompute_kernel<<<grid, FA_BLOCK, shm, stream>>>(
+      (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
+      (const float*)mask, (const float*)lse, (const float*)ddot, (short*)dq,
+      B, H, L, scale);
   return hipGetLastError();
 }
 

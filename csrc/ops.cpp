@@ -44,6 +44,10 @@ hipError_t flash_bwd_fused_launch(const void*, const void*, const void*,
                                   const void*, void*, void*, void*, int, int,
                                   int, float, hipStream_t);
 hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
+hipError_t flash_dq_recompute_launch(const void*, const void*, const void*,
+                                     const void*, const void*, const void*,
+                                     const void*, void*, int, int, int, float,
+                                     hipStream_t);
 hipError_t flash_dq_launch(const void*, const void*, void*, int, int, int,
                            hipStream_t);
 hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
@@ -367,7 +371,8 @@ std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v, torch::Tensor dout,
                                            c10::optional<torch::Tensor> mask,
                                            torch::Tensor lse,
-                                           torch::Tensor ddot, double scale) {
+                                           torch::Tensor ddot, double scale,
+                                           bool emit_ds) {
   check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
   check_bf16(dout, "dout"); check_f32(lse, "lse"); check_f32(ddot, "ddot");
   const long B = q.size(0), H = q.size(1), L = q.size(2);
@@ -377,16 +382,24 @@ std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
     check_f32(*mask, "mask");
     mptr = mask->data_ptr();
   }
-  auto ds = torch::empty({B, H, L, L}, q.options());
+  // emit_ds=false (the default path since round 2): no [B, H, L, L] dS
+  // materialization — dQ comes from flash_dq_recompute instead.
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
+  torch::Tensor ds;
+  void* ds_ptr = nullptr;
+  if (emit_ds) {
+    ds = torch::empty({B, H, L, L}, q.options());
+    ds_ptr = ds.data_ptr();
+  }
   CHECK_HIP(flash_bwd_fused_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                                    dout.data_ptr(), mptr, lse.data_ptr(),
-                                   ddot.data_ptr(), ds.data_ptr(),
+                                   ddot.data_ptr(), ds_ptr,
                                    dk.data_ptr(), dv.data_ptr(), (int)B,
                                    (int)H, (int)L, (float)scale,
                                    cur_stream()));
-  return {ds, dk, dv};
+  if (emit_ds) return {ds, dk, dv};
+  return {dk, dv};
 }
 
 torch::Tensor bias_grad(torch::Tensor dy) {
@@ -399,6 +412,29 @@ torch::Tensor bias_grad(torch::Tensor dy) {
   CHECK_HIP(colsum_bf16_launch(dy.data_ptr(), scratch.data_ptr(),
                                out.data_ptr(), N, D, cur_stream()));
   return out;
+}
+
+torch::Tensor flash_dq_recompute(torch::Tensor q, torch::Tensor k,
+                                 torch::Tensor v, torch::Tensor dout,
+                                 c10::optional<torch::Tensor> mask,
+                                 torch::Tensor lse, torch::Tensor ddot,
+                                 double scale) {
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  check_bf16(dout, "dout"); check_f32(lse, "lse"); check_f32(ddot, "ddot");
+  const long B = q.size(0), H = q.size(1), L = q.size(2);
+  TORCH_CHECK(q.size(3) == 64 && L % 32 == 0, "dh=64 and L%32==0 required");
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    mptr = mask->data_ptr();
+  }
+  auto dq = torch::empty_like(q);
+  CHECK_HIP(flash_dq_recompute_launch(q.data_ptr(), k.data_ptr(),
+                                      v.data_ptr(), dout.data_ptr(), mptr,
+                                      lse.data_ptr(), ddot.data_ptr(),
+                                      dq.data_ptr(), (int)B, (int)H, (int)L,
+                                      (float)scale, cur_stream()));
+  return dq;
 }
 
 torch::Tensor flash_dq(torch::Tensor ds, torch::Tensor k) {
@@ -461,7 +497,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("masked_pool_bwd", &masked_pool_bwd, "masked mean-pool bwd");
   m.def("flash_fwd", &flash_fwd, "flash attention fwd (gfx950 MFMA, dh=64)");
   m.def("flash_bwd_fused", &flash_bwd_fused,
-        "fused flash bwd: dS + register-accumulated dK/dV");
+        "fused flash bwd: register-accumulated dK/dV (+ optional dS)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("dout"),
+        py::arg("mask"), py::arg("lse"), py::arg("ddot"), py::arg("scale"),
+        py::arg("emit_ds") = false);
+  m.def("flash_dq_recompute", &flash_dq_recompute,
+        "dQ by recompute: S/dP/dS in-register, no dS materialization");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
   m.def("flash_dq", &flash_dq, "dQ = dS @ K (MFMA, tr_b16 K^T fragments)");
   m.def("bias_grad", &bias_grad, "bf16 column-sum for linear bias grads");

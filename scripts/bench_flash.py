@@ -29,36 +29,50 @@ def main():
     dt = (time.perf_counter() - t0) / iters
     flops = 4 * B * H * L * L * dh
     print(f"flash_fwd {dt*1e6:.1f} us/call  {flops/dt/1e12:.1f} TF")
-    # bwd stage
+    # bwd stage (round-2 default: dk/dv only, no dS materialization)
     do = torch.randn_like(q)
     ddot = ext.fa_dot(do, o)
     for _ in range(3):
-        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
+        dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
+        dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    print(f"flash_bwd_fused {dt*1e6:.1f} us/call  {4*flops/dt/1e12:.1f} TF-eq")
-    # whole backward chain incl. the custom dq kernel
+    print(f"flash_bwd_fused {dt*1e6:.1f} us/call  {3*flops/dt/1e12:.1f} TF-eq")
+    for _ in range(3):
+        dq = ext.flash_dq_recompute(q, k, v, do, mask, lse, ddot, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        dq = ext.flash_dq_recompute(q, k, v, do, mask, lse, ddot, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    print(f"flash_dq_recompute {dt*1e6:.1f} us/call  {3*flops/dt/1e12:.1f} TF-eq")
+    # round-1 chain for comparison (dS round-trip + dq-from-ds)
+    ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale,
+                                     emit_ds=True)
     for _ in range(3):
         dq = ext.flash_dq(ds, k)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
+        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale,
+                                         emit_ds=True)
         dq = ext.flash_dq(ds, k)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    print(f"flash_dq {dt*1e6:.1f} us/call  {flops/2/dt/1e12:.1f} TF")
+    print(f"r1 bwd chain (fused+emit_ds + dq) {dt*1e6:.1f} us")
+    del ds
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ds, dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
-        dq = ext.flash_dq(ds, k)
+        dk, dv = ext.flash_bwd_fused(q, k, v, do, mask, lse, ddot, scale)
+        dq = ext.flash_dq_recompute(q, k, v, do, mask, lse, ddot, scale)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    print(f"bwd chain (fused + dq) {dt*1e6:.1f} us")
+    print(f"r2 bwd chain (fused + dq_recompute) {dt*1e6:.1f} us")
 
 
 if __name__ == "__main__":

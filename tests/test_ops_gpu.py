@@ -268,7 +268,7 @@ def test_flash_bwd_fused_vs_reference():
     dd_ref = (do.float() * o.float()).sum(-1)
     assert torch.allclose(ddot, dd_ref, atol=2e-2, rtol=2e-2)
     dsg, dkg, dvg = ops.hip_ops().flash_bwd_fused(q, k, v, do, mg, lse,
-                                                  ddot, scale)
+                                                  ddot, scale, emit_ds=True)
     # reference
     s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale \
         + mask.cuda().view(B, 1, 1, L)
@@ -283,6 +283,15 @@ def test_flash_bwd_fused_vs_reference():
         (dvg.float() - dv_ref).abs().max()
     assert torch.allclose(dkg.float(), dk_ref, atol=5e-2, rtol=3e-2), \
         (dkg.float() - dk_ref).abs().max()
+    # default path returns dk/dv only (no dS materialization)
+    dk2, dv2 = ops.hip_ops().flash_bwd_fused(q, k, v, do, mg, lse,
+                                             ddot, scale)
+    assert torch.equal(dk2, dkg) and torch.equal(dv2, dvg)
+    # round-2 dQ-by-recompute kernel vs the fp32 reference
+    dqg = ops.hip_ops().flash_dq_recompute(q, k, v, do, mg, lse, ddot, scale)
+    dq_ref = torch.matmul(ds_ref, k.float())
+    assert torch.allclose(dqg.float(), dq_ref, atol=5e-2, rtol=3e-2), \
+        (dqg.float() - dq_ref).abs().max()
 
 
 def test_layernorm_bwd_with_residual_grad():
@@ -433,11 +442,16 @@ def test_kernels_bitwise_deterministic():
     o2, lse2 = ext.flash_fwd(q, k, v, mask, 0.125)
     assert torch.equal(o1, o2) and torch.equal(lse1, lse2)
     dd = ext.fa_dot(do, o1)
-    r1 = ext.flash_bwd_fused(q, k, v, do, mask, lse1, dd, 0.125)
-    r2 = ext.flash_bwd_fused(q, k, v, do, mask, lse1, dd, 0.125)
+    r1 = ext.flash_bwd_fused(q, k, v, do, mask, lse1, dd, 0.125,
+                             emit_ds=True)
+    r2 = ext.flash_bwd_fused(q, k, v, do, mask, lse1, dd, 0.125,
+                             emit_ds=True)
     for a, b in zip(r1, r2):
         assert torch.equal(a, b)
     assert torch.equal(ext.flash_dq(r1[0], k), ext.flash_dq(r1[0], k))
+    assert torch.equal(
+        ext.flash_dq_recompute(q, k, v, do, mask, lse1, dd, 0.125),
+        ext.flash_dq_recompute(q, k, v, do, mask, lse1, dd, 0.125))
     N, D = 4096, 1024
     x = _bf16(torch.randn(N, D))
     g = _bf16(torch.randn(D))

@@ -259,7 +259,9 @@ def lt_linear_gelu_bias(x, w1, b1):
 
 class _FlashAttentionFn(torch.autograd.Function):
     """Flash attention: MFMA forward (csrc/flash_attn.hip, O(L) memory, saves
-    logsumexp), fused recompute backward (dS + in-kernel dK/dV)."""
+    logsumexp); backward = two recompute kernels, O(L) memory end to end:
+    flash_bwd_fused (register-accumulated dK/dV) + flash_dq_recompute
+    (in-register S/dP/dS, accumulates dQ) — no [L, L] dS materialization."""
 
     @staticmethod
     def forward(ctx, q, k, v, mask, scale):
@@ -278,12 +280,13 @@ class _FlashAttentionFn(torch.autograd.Function):
         mask, scale = ctx.mask, ctx.scale
         do = do.contiguous()
         if q.is_cuda:
-            # fully-fused recompute: one MFMA kernel emits dS and
-            # register-accumulated dK/dV; dQ is the one remaining bmm
+            # two recompute MFMA kernels; dS never touches HBM (round 2 —
+            # the round-1 pipeline wrote + re-read a [B, H, L, L] bf16 dS)
             ddot = hip_ops().fa_dot(do, o)
-            ds, dk, dv = hip_ops().flash_bwd_fused(q, k, v, do, mask, lse,
-                                                   ddot, scale)
-            dq = hip_ops().flash_dq(ds, k)  # [q,kv] @ [kv,d], custom MFMA
+            dk, dv = hip_ops().flash_bwd_fused(q, k, v, do, mask, lse,
+                                               ddot, scale)
+            dq = hip_ops().flash_dq_recompute(q, k, v, do, mask, lse,
+                                              ddot, scale)
             return dq, dk, dv, None, None
         s = torch.matmul(q, k.transpose(-1, -2))
         p = reference.p_from_lse(s, mask, lse, scale)
