@@ -3,37 +3,57 @@
 // y = gelu(x + b); bwd recomputes the pre-activation from x,b (no saved
 // activation: 1 extra read instead of a [N,Dff] bf16 save — HBM3E-friendly).
 // dbias uses deterministic per-block LDS partials + colsum (layernorm.hip).
-// x: [N, D] bf16 row-major, b: [D] bf16.  Elementwise, memory-bound: 16 B/lane
-// vector loads, grid-stride, <= 2048 workgroups (guide G11/G13).
+// x: [N, D] bf16 row-major, b: [D] bf16.
+//
+// Round 2: 4 x b128 packets (64 B) per thread per iteration — the round-1
+// single-packet loop measured ~4.2 TB/s (one 16-B load in flight per
+// thread); issuing 4 back-to-back b128 loads per iteration raises
+// memory-level parallelism on the same grid.  Fast path needs the grid
+// stride a multiple of D (host rounds the grid), D % 32 == 0 and
+// n % 32 == 0 (both hold at every model shape; generic fallback otherwise).
 
 #include "common.h"
 
 #define BG_BLOCK 256
+#define BG_PK 4                      // packets of 8 bf16 per iteration
 
 extern "C" {
 
 __global__ void __launch_bounds__(BG_BLOCK)
 bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
                      short* __restrict__ y, long n_elem, int D) {
-  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
-  long stride = (long)gridDim.x * BG_BLOCK * 8;
-  if (stride % D == 0) {
-    // fixed column window: hoist the bias load and the 64-bit modulo
+  const int elems = 8 * BG_PK;
+  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * elems;
+  long stride = (long)gridDim.x * BG_BLOCK * elems;
+  if (stride % D == 0 && D % elems == 0 && n_elem % elems == 0) {
+    // fixed 32-column window per thread: hoist the bias loads
     const int col = (int)(idx0 % D);
-    short8_t b8 = *(const short8_t*)(b + col);
-    float bb[8];
+    float bb[elems];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) bb[j] = bf16_to_f32(b8[j]);
+    for (int p = 0; p < BG_PK; ++p) {
+      short8_t b8 = *(const short8_t*)(b + col + 8 * p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bb[8 * p + j] = bf16_to_f32(b8[j]);
+    }
     for (long i = idx0; i < n_elem; i += stride) {
-      short8_t v = *(const short8_t*)(x + i);
-      short8_t o;
+      short8_t v[BG_PK];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[j]) + bb[j]));
-      *(short8_t*)(y + i) = o;
+      for (int p = 0; p < BG_PK; ++p)
+        v[p] = *(const short8_t*)(x + i + 8 * p);
+#pragma unroll
+      for (int p = 0; p < BG_PK; ++p) {
+        short8_t o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[p][j]) + bb[8 * p + j]));
+        *(short8_t*)(y + i + 8 * p) = o;
+      }
     }
     return;
   }
+  // generic path (one packet per iteration, per-packet bias reload)
+  idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
+  stride = (long)gridDim.x * BG_BLOCK * 8;
   for (long i = idx0; i < n_elem; i += stride) {
     short8_t v = *(const short8_t*)(x + i);
     int col = (int)(i % D);  // D % 8 == 0 so the packet stays in one row
@@ -46,10 +66,9 @@ bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
   }
 }
 
-// When the grid stride is a multiple of D (the host picks such a grid), each
-// thread's packet stays on ONE fixed 8-column window for its whole loop, so
-// the dbias partial accumulates in 8 registers and costs 8 LDS atomics per
-// thread total, not 8 per element (was 8.1 ms/step, baseline profile).
+// dbias: each thread's fixed 32-column window accumulates in registers; 32
+// LDS atomics per thread at the END (not per element), then one per-block
+// f32 partial row for the deterministic colsum reduction.
 __global__ void __launch_bounds__(BG_BLOCK)
 bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
                      const short* __restrict__ b, short* __restrict__ dx,
@@ -58,32 +77,45 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   float* sdb = (float*)smem;  // [D]
   for (int i = threadIdx.x; i < D; i += BG_BLOCK) sdb[i] = 0.f;
   __syncthreads();
-  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
-  long stride = (long)gridDim.x * BG_BLOCK * 8;
-  if (stride % D == 0) {
-    float acc[8] = {0.f};
+  const int elems = 8 * BG_PK;
+  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * elems;
+  long stride = (long)gridDim.x * BG_BLOCK * elems;
+  if (stride % D == 0 && D % elems == 0 && n_elem % elems == 0) {
+    float acc[elems] = {0.f};
     const int col = (int)(idx0 % D);
-    short8_t bv = *(const short8_t*)(b + col);
-    float bb[8];
+    float bb[elems];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) bb[j] = bf16_to_f32(bv[j]);
+    for (int p = 0; p < BG_PK; ++p) {
+      short8_t b8 = *(const short8_t*)(b + col + 8 * p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bb[8 * p + j] = bf16_to_f32(b8[j]);
+    }
     for (long i = idx0; i < n_elem; i += stride) {
-      short8_t vd = *(const short8_t*)(dy + i);
-      short8_t vx = *(const short8_t*)(x + i);
-      short8_t o;
+      short8_t vd[BG_PK], vx[BG_PK];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = bf16_to_f32(vd[j]);
-        float pre = bf16_to_f32(vx[j]) + bb[j];
-        float dpre = g * gelu_tanh_grad(pre);
-        o[j] = f32_to_bf16(dpre);
-        acc[j] += dpre;  // dL/db = sum over rows of dy * gelu'(pre)
+      for (int p = 0; p < BG_PK; ++p) {
+        vd[p] = *(const short8_t*)(dy + i + 8 * p);
+        vx[p] = *(const short8_t*)(x + i + 8 * p);
       }
-      *(short8_t*)(dx + i) = o;
+#pragma unroll
+      for (int p = 0; p < BG_PK; ++p) {
+        short8_t o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = bf16_to_f32(vd[p][j]);
+          float pre = bf16_to_f32(vx[p][j]) + bb[8 * p + j];
+          float dpre = g * gelu_tanh_grad(pre);
+          o[j] = f32_to_bf16(dpre);
+          acc[8 * p + j] += dpre;
+        }
+        *(short8_t*)(dx + i + 8 * p) = o;
+      }
     }
 #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&sdb[col + j], acc[j]);
+    for (int j = 0; j < elems; ++j) atomicAdd(&sdb[col + j], acc[j]);
   } else {
+    idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
+    stride = (long)gridDim.x * BG_BLOCK * 8;
     for (long i = idx0; i < n_elem; i += stride) {
       short8_t vd = *(const short8_t*)(dy + i);
       short8_t vx = *(const short8_t*)(x + i);

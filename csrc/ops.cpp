@@ -132,7 +132,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(x);
   // bwd grid stays modest: the [rows, D] dgamma/dbeta workspace and its
   // column-sum scale linearly with the grid
-  int grid = (int)std::min<long>((N + 3) / 4, 1024);
+  int grid = (int)std::min<long>((N + 3) / 4, 2048);
   // the template path (D in {512, 1024}) writes ONE ws row per wave
   // (deterministic, no atomics); the general path writes one per block
   const bool templ = (D == 512 || D == 1024);
@@ -166,8 +166,10 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b) {
   const long n = x.numel();
   TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
   auto y = torch::empty_like(x);
-  int grid = (int)std::min<long>((n / 8 + 255) / 256, 2048);
-  long g0f = D / std::__gcd((long)D, 2048L);
+  // 32 elements per thread per iteration (BG_PK=4 in bias_gelu.hip); round
+  // the grid so the stride grid*256*32 is a multiple of D (fast path)
+  int grid = (int)std::min<long>((n / 32 + 255) / 256, 2048);
+  long g0f = D / std::__gcd((long)D, 8192L);
   if (g0f <= 2048) grid = (int)((grid + g0f - 1) / g0f * g0f);
   CHECK_HIP(bias_gelu_fwd_launch(x.data_ptr(), b.data_ptr(), y.data_ptr(), n,
                                  D, grid, cur_stream()));
@@ -180,10 +182,10 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   const int D = (int)x.size(-1);
   const long n = x.numel();
   auto dx = torch::empty_like(x);
-  int grid = (int)std::min<long>((n / 8 + 255) / 256, 1024);
-  // round the grid up to a multiple of D/gcd(D, 2048) so the kernel's grid
-  // stride is a multiple of D -> fixed per-thread column window (fast path)
-  long g = std::__gcd((long)D, 2048L);
+  int grid = (int)std::min<long>((n / 32 + 255) / 256, 1024);
+  // round the grid up to a multiple of D/gcd(D, 256*32) so the kernel's
+  // grid stride is a multiple of D -> fixed per-thread column window
+  long g = std::__gcd((long)D, 8192L);
   long g0 = D / g;
   if (g0 <= 2048) grid = (int)((grid + g0 - 1) / g0 * g0);
   auto ws = torch::empty({grid, D}, x.options().dtype(torch::kFloat32));
