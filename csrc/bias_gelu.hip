@@ -5,56 +5,62 @@
 // dbias uses deterministic per-block LDS partials + colsum (layernorm.hip).
 // x: [N, D] bf16 row-major, b: [D] bf16.
 //
-// Round 2: 4 x b128 packets (64 B) per thread per iteration — the round-1
-// single-packet loop measured ~4.2 TB/s (one 16-B load in flight per
-// thread); issuing 4 back-to-back b128 loads per iteration raises
-// memory-level parallelism on the same grid.  Fast path needs the grid
-// stride a multiple of D (host rounds the grid), D % 32 == 0 and
-// n % 32 == 0 (both hold at every model shape; generic fallback otherwise).
+// Round 2: 4 packets per thread per iteration for memory-level parallelism,
+// laid out BLOCK-STRIDED (packet p of thread t at tile + p*BG_BLOCK*8 + t*8)
+// so every b128 load instruction stays perfectly dense across the wave —
+// the first attempt used 4 CONSECUTIVE packets per thread (64 B lane
+// stride), which broke per-instruction coalescing and measured ~15% slower
+// than round 1.  Fast path needs the grid stride and the per-packet stride
+// (BG_BLOCK*8 = 2048) multiples of D and n % 8 == 0; generic fallback
+// otherwise (host rounds the grid accordingly).
 
 #include "common.h"
 
 #define BG_BLOCK 256
 #define BG_PK 4                      // packets of 8 bf16 per iteration
+#define BG_TILE (BG_BLOCK * 8)       // elements per packet-slab per block
 
 extern "C" {
 
 __global__ void __launch_bounds__(BG_BLOCK)
 bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
                      short* __restrict__ y, long n_elem, int D) {
-  const int elems = 8 * BG_PK;
-  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * elems;
-  long stride = (long)gridDim.x * BG_BLOCK * elems;
-  if (stride % D == 0 && D % elems == 0 && n_elem % elems == 0) {
-    // fixed 32-column window per thread: hoist the bias loads
-    const int col = (int)(idx0 % D);
-    float bb[elems];
+  long tile0 = (long)blockIdx.x * (BG_TILE * BG_PK) + threadIdx.x * 8;
+  long stride = (long)gridDim.x * (BG_TILE * BG_PK);
+  if (stride % D == 0) {
+    // per-packet columns are loop-invariant: hoist the bias loads
+    float bb[BG_PK * 8];
 #pragma unroll
     for (int p = 0; p < BG_PK; ++p) {
-      short8_t b8 = *(const short8_t*)(b + col + 8 * p);
+      int col = (int)((tile0 + p * BG_TILE) % D);
+      short8_t b8 = *(const short8_t*)(b + col);
 #pragma unroll
       for (int j = 0; j < 8; ++j) bb[8 * p + j] = bf16_to_f32(b8[j]);
     }
-    for (long i = idx0; i < n_elem; i += stride) {
+    for (long i = tile0; i < n_elem; i += stride) {
       short8_t v[BG_PK];
-#pragma unroll
-      for (int p = 0; p < BG_PK; ++p)
-        v[p] = *(const short8_t*)(x + i + 8 * p);
+      bool ok[BG_PK];
 #pragma unroll
       for (int p = 0; p < BG_PK; ++p) {
+        ok[p] = i + p * BG_TILE < n_elem;
+        v[p] = ok[p] ? *(const short8_t*)(x + i + p * BG_TILE) : short8_t{};
+      }
+#pragma unroll
+      for (int p = 0; p < BG_PK; ++p) {
+        if (!ok[p]) continue;
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           o[j] = f32_to_bf16(gelu_tanh(bf16_to_f32(v[p][j]) + bb[8 * p + j]));
-        *(short8_t*)(y + i + 8 * p) = o;
+        *(short8_t*)(y + i + p * BG_TILE) = o;
       }
     }
     return;
   }
   // generic path (one packet per iteration, per-packet bias reload)
-  idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
-  stride = (long)gridDim.x * BG_BLOCK * 8;
-  for (long i = idx0; i < n_elem; i += stride) {
+  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
+  long gstride = (long)gridDim.x * BG_BLOCK * 8;
+  for (long i = idx0; i < n_elem; i += gstride) {
     short8_t v = *(const short8_t*)(x + i);
     int col = (int)(i % D);  // D % 8 == 0 so the packet stays in one row
     short8_t b8 = *(const short8_t*)(b + col);
@@ -66,9 +72,9 @@ bias_gelu_fwd_kernel(const short* __restrict__ x, const short* __restrict__ b,
   }
 }
 
-// dbias: each thread's fixed 32-column window accumulates in registers; 32
-// LDS atomics per thread at the END (not per element), then one per-block
-// f32 partial row for the deterministic colsum reduction.
+// dbias: each thread's per-packet fixed 8-column windows accumulate in
+// registers; 32 LDS atomics per thread at the END (not per element), then
+// one per-block f32 partial row for the deterministic colsum reduction.
 __global__ void __launch_bounds__(BG_BLOCK)
 bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
                      const short* __restrict__ b, short* __restrict__ dx,
@@ -77,28 +83,31 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   float* sdb = (float*)smem;  // [D]
   for (int i = threadIdx.x; i < D; i += BG_BLOCK) sdb[i] = 0.f;
   __syncthreads();
-  const int elems = 8 * BG_PK;
-  long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * elems;
-  long stride = (long)gridDim.x * BG_BLOCK * elems;
-  if (stride % D == 0 && D % elems == 0 && n_elem % elems == 0) {
-    float acc[elems] = {0.f};
-    const int col = (int)(idx0 % D);
-    float bb[elems];
+  long tile0 = (long)blockIdx.x * (BG_TILE * BG_PK) + threadIdx.x * 8;
+  long stride = (long)gridDim.x * (BG_TILE * BG_PK);
+  if (stride % D == 0) {
+    float acc[BG_PK * 8] = {0.f};
+    int cols[BG_PK];
+    float bb[BG_PK * 8];
 #pragma unroll
     for (int p = 0; p < BG_PK; ++p) {
-      short8_t b8 = *(const short8_t*)(b + col + 8 * p);
+      cols[p] = (int)((tile0 + p * BG_TILE) % D);
+      short8_t b8 = *(const short8_t*)(b + cols[p]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) bb[8 * p + j] = bf16_to_f32(b8[j]);
     }
-    for (long i = idx0; i < n_elem; i += stride) {
+    for (long i = tile0; i < n_elem; i += stride) {
       short8_t vd[BG_PK], vx[BG_PK];
+      bool ok[BG_PK];
 #pragma unroll
       for (int p = 0; p < BG_PK; ++p) {
-        vd[p] = *(const short8_t*)(dy + i + 8 * p);
-        vx[p] = *(const short8_t*)(x + i + 8 * p);
+        ok[p] = i + p * BG_TILE < n_elem;
+        vd[p] = ok[p] ? *(const short8_t*)(dy + i + p * BG_TILE) : short8_t{};
+        vx[p] = ok[p] ? *(const short8_t*)(x + i + p * BG_TILE) : short8_t{};
       }
 #pragma unroll
       for (int p = 0; p < BG_PK; ++p) {
+        if (!ok[p]) continue;
         short8_t o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -108,15 +117,17 @@ bias_gelu_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ x,
           o[j] = f32_to_bf16(dpre);
           acc[8 * p + j] += dpre;
         }
-        *(short8_t*)(dx + i + 8 * p) = o;
+        *(short8_t*)(dx + i + p * BG_TILE) = o;
       }
     }
 #pragma unroll
-    for (int j = 0; j < elems; ++j) atomicAdd(&sdb[col + j], acc[j]);
+    for (int p = 0; p < BG_PK; ++p)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&sdb[cols[p] + j], acc[8 * p + j]);
   } else {
-    idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
-    stride = (long)gridDim.x * BG_BLOCK * 8;
-    for (long i = idx0; i < n_elem; i += stride) {
+    long idx0 = ((long)blockIdx.x * BG_BLOCK + threadIdx.x) * 8;
+    long gstride = (long)gridDim.x * BG_BLOCK * 8;
+    for (long i = idx0; i < n_elem; i += gstride) {
       short8_t vd = *(const short8_t*)(dy + i);
       short8_t vx = *(const short8_t*)(x + i);
       int col = (int)(i % D);

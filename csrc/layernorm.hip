@@ -330,6 +330,141 @@ ln_bwd_t(const short* __restrict__ dy, const short* __restrict__ x,
   }
 }
 
+// Two rows per wave (round 2): the single-row ln_bwd_t is latency-bound on
+// its serial chain (2 dependent wave_sum shuffle reductions + the dx store
+// per row; measured ~2-3 TB/s vs ln_fwd's 6.4).  Interleaving two
+// independent rows per wave doubles the work inside the same latency
+// shadow — the four wave_sum chains (s1/s2 x 2 rows) issue back-to-back.
+// dgamma/dbeta accumulators are shared (summed over both rows), so the
+// per-wave ws row layout and the deterministic colsum are unchanged.
+template <int PKTS, bool HASDE>
+__global__ void __launch_bounds__(BLOCK)
+ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
+          const short* __restrict__ gamma, const float* __restrict__ mean_in,
+          const float* __restrict__ rstd_in,
+          const short* __restrict__ ds_extra, short* __restrict__ dx,
+          float* __restrict__ ws_dgamma, float* __restrict__ ws_dbeta,
+          int N, int D) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  float gv[PKTS * 8];
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    short8_t g8 = *(const short8_t*)(gamma + (p * WAVE + lane) * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) gv[p * 8 + j] = bf16_to_f32(g8[j]);
+  }
+  float dg_acc[PKTS * 8] = {0.f}, db_acc[PKTS * 8] = {0.f};
+  const int rstride = gridDim.x * WAVES_PER_BLOCK * 2;
+  int rowA = (blockIdx.x * WAVES_PER_BLOCK + wid) * 2;
+  short8_t vdA[PKTS], vxA[PKTS], veA[PKTS];
+  short8_t vdB[PKTS], vxB[PKTS], veB[PKTS];
+  bool actA = rowA < N, actB = rowA + 1 < N;
+  if (actA) {
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      int base = (p * WAVE + lane) * 8;
+      vdA[p] = *(const short8_t*)(dy + (long)rowA * D + base);
+      vxA[p] = *(const short8_t*)(x + (long)rowA * D + base);
+      if (HASDE) veA[p] = *(const short8_t*)(ds_extra + (long)rowA * D + base);
+      if (actB) {
+        vdB[p] = *(const short8_t*)(dy + (long)(rowA + 1) * D + base);
+        vxB[p] = *(const short8_t*)(x + (long)(rowA + 1) * D + base);
+        if (HASDE)
+          veB[p] = *(const short8_t*)(ds_extra + (long)(rowA + 1) * D + base);
+      }
+    }
+  }
+  while (actA) {
+    const int next = rowA + rstride;
+    const float meanA = mean_in[rowA], rstdA = rstd_in[rowA];
+    const float meanB = actB ? mean_in[rowA + 1] : 0.f;
+    const float rstdB = actB ? rstd_in[rowA + 1] : 0.f;
+    float xhA[PKTS * 8], dygA[PKTS * 8], evA[PKTS * 8];
+    float xhB[PKTS * 8], dygB[PKTS * 8], evB[PKTS * 8];
+    float s1A = 0.f, s2A = 0.f, s1B = 0.f, s2B = 0.f;
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = p * 8 + j;
+        float dA = bf16_to_f32(vdA[p][j]);
+        float hA = (bf16_to_f32(vxA[p][j]) - meanA) * rstdA;
+        float gA = dA * gv[k];
+        if (HASDE) evA[k] = bf16_to_f32(veA[p][j]);
+        xhA[k] = hA; dygA[k] = gA;
+        dg_acc[k] += dA * hA;
+        db_acc[k] += dA;
+        s1A += gA; s2A += gA * hA;
+        if (actB) {
+          float dB = bf16_to_f32(vdB[p][j]);
+          float hB = (bf16_to_f32(vxB[p][j]) - meanB) * rstdB;
+          float gB = dB * gv[k];
+          if (HASDE) evB[k] = bf16_to_f32(veB[p][j]);
+          xhB[k] = hB; dygB[k] = gB;
+          dg_acc[k] += dB * hB;
+          db_acc[k] += dB;
+          s1B += gB; s2B += gB * hB;
+        }
+      }
+    }
+    if (next < N) {
+#pragma unroll
+      for (int p = 0; p < PKTS; ++p) {
+        int base = (p * WAVE + lane) * 8;
+        vdA[p] = *(const short8_t*)(dy + (long)next * D + base);
+        vxA[p] = *(const short8_t*)(x + (long)next * D + base);
+        if (HASDE) veA[p] = *(const short8_t*)(ds_extra + (long)next * D + base);
+        if (next + 1 < N) {
+          vdB[p] = *(const short8_t*)(dy + (long)(next + 1) * D + base);
+          vxB[p] = *(const short8_t*)(x + (long)(next + 1) * D + base);
+          if (HASDE)
+            veB[p] = *(const short8_t*)(ds_extra + (long)(next + 1) * D + base);
+        }
+      }
+    }
+    s1A = wave_sum(s1A) / (float)D;
+    s2A = wave_sum(s2A) / (float)D;
+    if (actB) {
+      s1B = wave_sum(s1B) / (float)D;
+      s2B = wave_sum(s2B) / (float)D;
+    }
+#pragma unroll
+    for (int p = 0; p < PKTS; ++p) {
+      short8_t oA, oB;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = p * 8 + j;
+        float dA_ = rstdA * (dygA[k] - s1A - xhA[k] * s2A);
+        if (HASDE) dA_ += evA[k];
+        oA[j] = f32_to_bf16(dA_);
+        if (actB) {
+          float dB_ = rstdB * (dygB[k] - s1B - xhB[k] * s2B);
+          if (HASDE) dB_ += evB[k];
+          oB[j] = f32_to_bf16(dB_);
+        }
+      }
+      int base = (p * WAVE + lane) * 8;
+      *(short8_t*)(dx + (long)rowA * D + base) = oA;
+      if (actB) *(short8_t*)(dx + (long)(rowA + 1) * D + base) = oB;
+    }
+    if (next >= N) break;
+    rowA = next;
+    actB = rowA + 1 < N;
+  }
+  float* og = ws_dgamma + (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
+  float* ob = ws_dbeta + (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
+#pragma unroll
+  for (int p = 0; p < PKTS; ++p) {
+    int base = (p * WAVE + lane) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      og[base + j] = dg_acc[p * 8 + j];
+      ob[base + j] = db_acc[p * 8 + j];
+    }
+  }
+}
+
 extern "C" {
 
 // dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg * xhat)),  dyg = dy * gamma
@@ -575,12 +710,12 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
 #define LNB_T(P)                                                              \
   do {                                                                        \
     if (ds_extra)                                                             \
-      ln_bwd_t<P, true><<<grid, BLOCK, 0, stream>>>(                          \
+      ln_bwd_t2<P, true><<<grid, BLOCK, 0, stream>>>(                         \
           (const short*)dy, (const short*)x, (const short*)gamma,             \
           (const float*)mean, (const float*)rstd, (const short*)ds_extra,     \
           (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);             \
     else                                                                      \
-      ln_bwd_t<P, false><<<grid, BLOCK, 0, stream>>>(                         \
+      ln_bwd_t2<P, false><<<grid, BLOCK, 0, stream>>>(                        \
           (const short*)dy, (const short*)x, (const short*)gamma,             \
           (const float*)mean, (const float*)rstd, nullptr, (short*)dx,        \
           (float*)ws_dgamma, (float*)ws_dbeta, N, D);                         \

@@ -132,7 +132,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(x);
   // bwd grid stays modest: the [rows, D] dgamma/dbeta workspace and its
   // column-sum scale linearly with the grid
-  int grid = (int)std::min<long>((N + 3) / 4, 2048);
+  int grid = (int)std::min<long>((N + 3) / 4, 1024);
   // the template path (D in {512, 1024}) writes ONE ws row per wave
   // (deterministic, no atomics); the general path writes one per block
   const bool templ = (D == 512 || D == 1024);
