@@ -11,6 +11,8 @@
 // lane (transformer encoder LayerNorm).
 
 #include "common.h"
+#include <cstdlib>
+#include <cstring>
 
 #define WAVES_PER_BLOCK 4
 #define BLOCK (WAVES_PER_BLOCK * WAVE)
@@ -380,8 +382,8 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
     const float meanA = mean_in[rowA], rstdA = rstd_in[rowA];
     const float meanB = actB ? mean_in[rowA + 1] : 0.f;
     const float rstdB = actB ? rstd_in[rowA + 1] : 0.f;
-    float xhA[PKTS * 8], dygA[PKTS * 8], evA[PKTS * 8];
-    float xhB[PKTS * 8], dygB[PKTS * 8], evB[PKTS * 8];
+    float xhA[PKTS * 8], dygA[PKTS * 8];
+    float xhB[PKTS * 8], dygB[PKTS * 8];
     float s1A = 0.f, s2A = 0.f, s1B = 0.f, s2B = 0.f;
 #pragma unroll
     for (int p = 0; p < PKTS; ++p) {
@@ -391,7 +393,6 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
         float dA = bf16_to_f32(vdA[p][j]);
         float hA = (bf16_to_f32(vxA[p][j]) - meanA) * rstdA;
         float gA = dA * gv[k];
-        if (HASDE) evA[k] = bf16_to_f32(veA[p][j]);
         xhA[k] = hA; dygA[k] = gA;
         dg_acc[k] += dA * hA;
         db_acc[k] += dA;
@@ -400,7 +401,6 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
           float dB = bf16_to_f32(vdB[p][j]);
           float hB = (bf16_to_f32(vxB[p][j]) - meanB) * rstdB;
           float gB = dB * gv[k];
-          if (HASDE) evB[k] = bf16_to_f32(veB[p][j]);
           xhB[k] = hB; dygB[k] = gB;
           dg_acc[k] += dB * hB;
           db_acc[k] += dB;
@@ -414,12 +414,9 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
         int base = (p * WAVE + lane) * 8;
         vdA[p] = *(const short8_t*)(dy + (long)next * D + base);
         vxA[p] = *(const short8_t*)(x + (long)next * D + base);
-        if (HASDE) veA[p] = *(const short8_t*)(ds_extra + (long)next * D + base);
         if (next + 1 < N) {
           vdB[p] = *(const short8_t*)(dy + (long)(next + 1) * D + base);
           vxB[p] = *(const short8_t*)(x + (long)(next + 1) * D + base);
-          if (HASDE)
-            veB[p] = *(const short8_t*)(ds_extra + (long)(next + 1) * D + base);
         }
       }
     }
@@ -436,11 +433,11 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
       for (int j = 0; j < 8; ++j) {
         int k = p * 8 + j;
         float dA_ = rstdA * (dygA[k] - s1A - xhA[k] * s2A);
-        if (HASDE) dA_ += evA[k];
+        if (HASDE) dA_ += bf16_to_f32(veA[p][j]);
         oA[j] = f32_to_bf16(dA_);
         if (actB) {
           float dB_ = rstdB * (dygB[k] - s1B - xhB[k] * s2B);
-          if (HASDE) dB_ += evB[k];
+          if (HASDE) dB_ += bf16_to_f32(veB[p][j]);
           oB[j] = f32_to_bf16(dB_);
         }
       }
@@ -451,6 +448,15 @@ ln_bwd_t2(const short* __restrict__ dy, const short* __restrict__ x,
     if (next >= N) break;
     rowA = next;
     actB = rowA + 1 < N;
+    if (HASDE) {
+#pragma unroll
+      for (int p = 0; p < PKTS; ++p) {
+        int base = (p * WAVE + lane) * 8;
+        veA[p] = *(const short8_t*)(ds_extra + (long)rowA * D + base);
+        if (actB)
+          veB[p] = *(const short8_t*)(ds_extra + (long)(rowA + 1) * D + base);
+      }
+    }
   }
   float* og = ws_dgamma + (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
   float* ob = ws_dbeta + (long)(blockIdx.x * WAVES_PER_BLOCK + wid) * D;
@@ -701,6 +707,16 @@ hipError_t ln_fwd_launch(const void* x, const void* res, const void* gamma,
   return hipGetLastError();
 }
 
+// runtime A/B: TOSEM_LN_BWD=1row selects the single-row ln_bwd_t variant
+static int ln_bwd_use_1row() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TOSEM_LN_BWD");
+    v = (e && strcmp(e, "1row") == 0) ? 1 : 0;
+  }
+  return v;
+}
+
 hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
                          const void* mean, const void* rstd,
                          const void* ds_extra, void* dx,
@@ -709,6 +725,19 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
   size_t shm = (size_t)D * 2 * sizeof(float);
 #define LNB_T(P)                                                              \
   do {                                                                        \
+    if (ln_bwd_use_1row()) {                                                  \
+      if (ds_extra)                                                           \
+        ln_bwd_t<P, true><<<grid, BLOCK, 0, stream>>>(                        \
+            (const short*)dy, (const short*)x, (const short*)gamma,           \
+            (const float*)mean, (const float*)rstd, (const short*)ds_extra,   \
+            (short*)dx, (float*)ws_dgamma, (float*)ws_dbeta, N, D);           \
+      else                                                                    \
+        ln_bwd_t<P, false><<<grid, BLOCK, 0, stream>>>(                       \
+            (const short*)dy, (const short*)x, (const short*)gamma,           \
+            (const float*)mean, (const float*)rstd, nullptr, (short*)dx,      \
+            (float*)ws_dgamma, (float*)ws_dbeta, N, D);                       \
+      break;                                                                  \
+    }                                                                         \
     if (ds_extra)                                                             \
       ln_bwd_t2<P, true><<<grid, BLOCK, 0, stream>>>(                         \
           (const short*)dy, (const short*)x, (const short*)gamma,             \

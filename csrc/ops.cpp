@@ -168,7 +168,7 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b) {
   auto y = torch::empty_like(x);
   // 32 elements per thread per iteration (BG_PK=4 in bias_gelu.hip); round
   // the grid so the stride grid*256*32 is a multiple of D (fast path)
-  int grid = (int)std::min<long>((n / 32 + 255) / 256, 2048);
+  int grid = (int)std::min<long>((n / 32 + 255) / 256, 4096);
   long g0f = D / std::__gcd((long)D, 8192L);
   if (g0f <= 2048) grid = (int)((grid + g0f - 1) / g0f * g0f);
   CHECK_HIP(bias_gelu_fwd_launch(x.data_ptr(), b.data_ptr(), y.data_ptr(), n,
