@@ -43,11 +43,22 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
+// Tensor geometry (round 2): the kernels address q/k/v rows as
+//   base + b*qkv_bs + h*qkv_hs + l*qkv_rs        (64 contiguous shorts)
+// and o/dout rows as b*o_bs + h*o_hs + l*o_rs — so the SAME kernels run on
+// the bmm-style [B, H, L, 64] layout (qkv_bs=H*L*64, qkv_hs=L*64,
+// qkv_rs=64) and directly on the packed QKV-projection output [B, L, 3D]
+// (qkv_bs=L*3D, qkv_hs=64, qkv_rs=3D, k/v base-offset D/2D) with the
+// attention output written straight into the [B, L, D] proj input
+// (o_bs=L*D, o_hs=64, o_rs=D).  The packed path deletes the four
+// qkv/out repack kernels (+their backward merges) from the model step.
 extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
                  short* __restrict__ o, float* __restrict__ lse,
-                 int B, int H, int L, float scale) {
+                 int B, int H, int L, float scale,
+                 long qkv_bs, long qkv_hs, int qkv_rs,
+                 long o_bs, long o_hs, int o_rs) {
   // Each wave processes TWO independent 32-row q-blocks against the shared
   // K/V tile: the second block's MFMAs overlap the first block's serial
   // softmax chain (ILP within the wave).  2 waves/SIMD by registers; the
@@ -69,7 +80,9 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   int bh = bid / n_qblocks;
   int qb = bid % n_qblocks;
   const int b = bh / H;
-  const long bh_off = (long)bh * L * FA_DH;
+  const int h = bh % H;
+  const long qkv_off = (long)b * qkv_bs + (long)h * qkv_hs;
+  const long o_off = (long)b * o_bs + (long)h * o_hs;
   // wave's two q-blocks: rows [qA, qA+32) and [qB, qB+32)
   const int q_baseA = qb * rows_per_wg + wid * FA_QB;
   const int q_baseB = q_baseA + FA_QWG;
@@ -81,8 +94,8 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
 
   short8_t qfA[4], qfB[4];
   {
-    const short* qrA = q + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
-    const short* qrB = q + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
+    const short* qrA = q + qkv_off + (long)(validA ? my_qA : L - 1) * qkv_rs;
+    const short* qrB = q + qkv_off + (long)(validB ? my_qB : L - 1) * qkv_rs;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       qfA[c] = *(const short8_t*)(qrA + c * 16 + half * 8);
@@ -97,9 +110,9 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
 
   const int n_kv = L / FA_KVB;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
-  short8_t kv8 = *(const short8_t*)(k + bh_off + (long)srow * FA_DH +
+  short8_t kv8 = *(const short8_t*)(k + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
-  short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
+  short8_t vv8 = *(const short8_t*)(v + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
@@ -110,10 +123,10 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
     __syncthreads();
     if (kt + 1 < n_kv) {
-      kv8 = *(const short8_t*)(k + bh_off +
-                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
-      vv8 = *(const short8_t*)(v + bh_off +
-                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+      kv8 = *(const short8_t*)(k + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
     }
 
     // ---- QK^T for BOTH q-blocks (8 back-to-back MFMAs) ----
@@ -271,10 +284,10 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
       int qA = q_baseA + rloc, qB = q_baseB + rloc;
       if (qA < L)
-        o[bh_off + (long)qA * FA_DH + 32 * t + col] =
+        o[o_off + (long)qA * o_rs + 32 * t + col] =
             f32_to_bf16(oA[t][r] * alpha_lds[wid * 64 + rloc]);
       if (qB < L)
-        o[bh_off + (long)qB * FA_DH + 32 * t + col] =
+        o[o_off + (long)qB * o_rs + 32 * t + col] =
             f32_to_bf16(oB[t][r] * alpha_lds[wid * 64 + 32 + rloc]);
     }
   }
@@ -290,7 +303,27 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
   size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
-      (short*)o, (float*)lse, B, H, L, scale);
+      (short*)o, (float*)lse, B, H, L, scale,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH);
+  return hipGetLastError();
+}
+
+// qkv: [B, L, 3D] packed projection output (D = H*64); o: [B, L, D]
+extern "C" hipError_t flash_fwd_packed_launch(const void* qkv,
+                                              const void* mask, void* o,
+                                              void* lse, int B, int H, int L,
+                                              float scale,
+                                              hipStream_t stream) {
+  const int D = H * FA_DH;
+  int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
+  dim3 grid(B * H * n_qblocks);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
+  flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
+      (const short*)qkv, (const short*)qkv + D, (const short*)qkv + 2 * D,
+      (const float*)mask, (short*)o, (float*)lse, B, H, L, scale,
+      (long)L * 3 * D, 64L, 3 * D,
+      (long)L * D, 64L, D);
   return hipGetLastError();
 }
 
@@ -331,7 +364,9 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
                        const float* __restrict__ ddot,
                        short* __restrict__ ds, short* __restrict__ dk,
                        short* __restrict__ dv,
-                       int B, int H, int L, float scale) {
+                       int B, int H, int L, float scale,
+                       long qkv_bs, long qkv_hs, int qkv_rs,
+                       long o_bs, long o_hs, int o_rs) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = (short*)smem;                        // swizzled [32][64]
   short* do_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
@@ -349,7 +384,9 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
   int bh = bid / n_kvblocks;
   int kb = bid % n_kvblocks;
   const int b = bh / H;
-  const long bh_off = (long)bh * L * FA_DH;
+  const int h = bh % H;
+  const long qkv_off = (long)b * qkv_bs + (long)h * qkv_hs;
+  const long o_off = (long)b * o_bs + (long)h * o_hs;
   const long bh_sq = (long)bh * L * L;
   const int kv_base = kb * FA_QWG + wid * FA_KVB;     // this wave's 32 kv rows
   const int my_kv = kv_base + col;
@@ -360,8 +397,8 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // K and V fragments for this wave's kv block (resident all kernel)
   short8_t kf[4], vf[4];
   {
-    const short* kr = k + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
-    const short* vr = v + bh_off + (long)(kv_valid ? my_kv : L - 1) * FA_DH;
+    const short* kr = k + qkv_off + (long)(kv_valid ? my_kv : L - 1) * qkv_rs;
+    const short* vr = v + qkv_off + (long)(kv_valid ? my_kv : L - 1) * qkv_rs;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       kf[c] = *(const short8_t*)(kr + c * 16 + half * 8);
@@ -377,9 +414,9 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
 
   const int n_q = L / 32;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
-  short8_t qv8 = *(const short8_t*)(q + bh_off + (long)srow * FA_DH +
+  short8_t qv8 = *(const short8_t*)(q + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
-  short8_t dv8 = *(const short8_t*)(dout + bh_off + (long)srow * FA_DH +
+  short8_t dv8 = *(const short8_t*)(dout + o_off + (long)srow * o_rs +
                                     (sc8 >> 1));
   for (int qt = 0; qt < n_q; ++qt) {
     const int q0 = qt * 32;
@@ -394,9 +431,9 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
     __syncthreads();
     if (qt + 1 < n_q) {
-      qv8 = *(const short8_t*)(q + bh_off + (long)(q0 + 32 + srow) * FA_DH +
+      qv8 = *(const short8_t*)(q + qkv_off + (long)(q0 + 32 + srow) * qkv_rs +
                                (sc8 >> 1));
-      dv8 = *(const short8_t*)(dout + bh_off + (long)(q0 + 32 + srow) * FA_DH +
+      dv8 = *(const short8_t*)(dout + o_off + (long)(q0 + 32 + srow) * o_rs +
                                (sc8 >> 1));
     }
 
@@ -554,7 +591,7 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
     for (int r = 0; r < 16; ++r) {
       int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
       if (kv_base + kvl >= L) continue;
-      long off = bh_off + (long)(kv_base + kvl) * FA_DH + 32 * t + col;
+      long off = qkv_off + (long)(kv_base + kvl) * qkv_rs + 32 * t + col;
       dv[off] = f32_to_bf16(dv_acc[t][r]);
       dk[off] = f32_to_bf16(dk_acc[t][r]);
     }
@@ -571,7 +608,28 @@ extern "C" hipError_t flash_bwd_fused_launch(
   flash_bwd_fused_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)mask, (const float*)lse, (const float*)ddot, (short*)ds,
-      (short*)dk, (short*)dv, B, H, L, scale);
+      (short*)dk, (short*)dv, B, H, L, scale,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH);
+  return hipGetLastError();
+}
+
+// packed: qkv/dqkv [B, L, 3D], dout [B, L, D]; dk/dv land inside dqkv
+extern "C" hipError_t flash_bwd_fused_packed_launch(
+    const void* qkv, const void* dout, const void* mask, const void* lse,
+    const void* ddot, void* dqkv, int B, int H, int L, float scale,
+    hipStream_t stream) {
+  const int D = H * FA_DH;
+  int n_kvblocks = (L + FA_QWG - 1) / FA_QWG;
+  dim3 grid(B * H * n_kvblocks);
+  size_t shm = 2 * K_LDS_BYTES + 64 * sizeof(float);
+  flash_bwd_fused_kernel<<<grid, FA_BLOCK, shm, stream>>>(
+      (const short*)qkv, (const short*)qkv + D, (const short*)qkv + 2 * D,
+      (const short*)dout, (const float*)mask, (const float*)lse,
+      (const float*)ddot, nullptr,
+      (short*)dqkv + D, (short*)dqkv + 2 * D, B, H, L, scale,
+      (long)L * 3 * D, 64L, 3 * D,
+      (long)L * D, 64L, D);
   return hipGetLastError();
 }
 
@@ -740,7 +798,9 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
                           const float* __restrict__ lse,
                           const float* __restrict__ ddot,
                           short* __restrict__ dq,
-                          int B, int H, int L, float scale) {
+                          int B, int H, int L, float scale,
+                          long qkv_bs, long qkv_hs, int qkv_rs,
+                          long o_bs, long o_hs, int o_rs) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = (short*)smem;                       // swizzled [32][64]
   short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
@@ -757,7 +817,9 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
   int bh = bid / n_qblocks;
   int qb = bid % n_qblocks;
   const int b = bh / H;
-  const long bh_off = (long)bh * L * FA_DH;
+  const int h = bh % H;
+  const long qkv_off = (long)b * qkv_bs + (long)h * qkv_hs;
+  const long o_off = (long)b * o_bs + (long)h * o_hs;
   const int q_baseA = qb * rows_per_wg + wid * FA_QB;
   const int q_baseB = q_baseA + FA_QWG;
   const int my_qA = q_baseA + col;
@@ -768,10 +830,10 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
 
   short8_t qfA[4], qfB[4], dofA[4], dofB[4];
   {
-    const short* qrA = q + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
-    const short* qrB = q + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
-    const short* drA = dout + bh_off + (long)(validA ? my_qA : L - 1) * FA_DH;
-    const short* drB = dout + bh_off + (long)(validB ? my_qB : L - 1) * FA_DH;
+    const short* qrA = q + qkv_off + (long)(validA ? my_qA : L - 1) * qkv_rs;
+    const short* qrB = q + qkv_off + (long)(validB ? my_qB : L - 1) * qkv_rs;
+    const short* drA = dout + o_off + (long)(validA ? my_qA : L - 1) * o_rs;
+    const short* drB = dout + o_off + (long)(validB ? my_qB : L - 1) * o_rs;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       qfA[c] = *(const short8_t*)(qrA + c * 16 + half * 8);
@@ -791,9 +853,9 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
 
   const int n_kv = L / FA_KVB;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
-  short8_t kv8 = *(const short8_t*)(k + bh_off + (long)srow * FA_DH +
+  short8_t kv8 = *(const short8_t*)(k + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
-  short8_t vv8 = *(const short8_t*)(v + bh_off + (long)srow * FA_DH +
+  short8_t vv8 = *(const short8_t*)(v + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
   for (int kt = 0; kt < n_kv; ++kt) {
     const int kv0 = kt * FA_KVB;
@@ -804,10 +866,10 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
     }
     __syncthreads();
     if (kt + 1 < n_kv) {
-      kv8 = *(const short8_t*)(k + bh_off +
-                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
-      vv8 = *(const short8_t*)(v + bh_off +
-                               (long)(kv0 + FA_KVB + srow) * FA_DH + (sc8 >> 1));
+      kv8 = *(const short8_t*)(k + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
     }
 
     // ---- S and dP for BOTH q-blocks (16 back-to-back MFMAs) ----
@@ -933,9 +995,11 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
       int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
       int qA = q_baseA + rloc, qB = q_baseB + rloc;
       if (qA < L)
-        dq[bh_off + (long)qA * FA_DH + 32 * t + col] = f32_to_bf16(dqA[t][r]);
+        dq[qkv_off + (long)qA * qkv_rs + 32 * t + col] =
+            f32_to_bf16(dqA[t][r]);
       if (qB < L)
-        dq[bh_off + (long)qB * FA_DH + 32 * t + col] = f32_to_bf16(dqB[t][r]);
+        dq[qkv_off + (long)qB * qkv_rs + 32 * t + col] =
+            f32_to_bf16(dqB[t][r]);
     }
   }
 }
@@ -950,7 +1014,27 @@ extern "C" hipError_t flash_dq_recompute_launch(
   flash_dq_recompute_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)mask, (const float*)lse, (const float*)ddot, (short*)dq,
-      B, H, L, scale);
+      B, H, L, scale,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH,
+      (long)H * L * FA_DH, (long)L * FA_DH, FA_DH);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t flash_dq_recompute_packed_launch(
+    const void* qkv, const void* dout, const void* mask, const void* lse,
+    const void* ddot, void* dqkv, int B, int H, int L, float scale,
+    hipStream_t stream) {
+  const int D = H * FA_DH;
+  int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
+  dim3 grid(B * H * n_qblocks);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES;
+  flash_dq_recompute_kernel<<<grid, FA_BLOCK, shm, stream>>>(
+      (const short*)qkv, (const short*)qkv + D, (const short*)qkv + 2 * D,
+      (const short*)dout, (const float*)mask, (const float*)lse,
+      (const float*)ddot, (short*)dqkv,
+      B, H, L, scale,
+      (long)L * 3 * D, 64L, 3 * D,
+      (long)L * D, 64L, D);
   return hipGetLastError();
 }
 
@@ -972,6 +1056,42 @@ fa_dot_kernel(const short* __restrict__ dout, const short* __restrict__ o,
 #pragma unroll
   for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 16);
   if (sub == 0) ddot[row] = s;
+}
+
+// packed fa_dot: dout/o are [B, L, D] (D = H*64); ddot is [B, H, L].
+extern "C" __global__ void __launch_bounds__(256)
+fa_dot_packed_kernel(const short* __restrict__ dout,
+                     const short* __restrict__ o, float* __restrict__ ddot,
+                     int B, int H, int L) {
+  const int D = H * FA_DH;
+  long n_rows = (long)B * L * H;          // one 64-elem chunk per (b, l, h)
+  long chunk = ((long)blockIdx.x * 256 + threadIdx.x) >> 4;
+  if (chunk >= n_rows) return;
+  int sub = threadIdx.x & 15;
+  int h = (int)(chunk % H);
+  long bl = chunk / H;                     // b * L + l
+  int l = (int)(bl % L);
+  int b = (int)(bl / L);
+  const short* dr = dout + bl * D + h * FA_DH + sub * 4;
+  const short* orow = o + bl * D + h * FA_DH + sub * 4;
+  short4_t a = *(const short4_t*)dr;
+  short4_t c = *(const short4_t*)orow;
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) s += bf16_to_f32(a[j]) * bf16_to_f32(c[j]);
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 16);
+  if (sub == 0) ddot[((long)b * H + h) * L + l] = s;
+}
+
+extern "C" hipError_t fa_dot_packed_launch(const void* dout, const void* o,
+                                           void* ddot, int B, int H, int L,
+                                           hipStream_t stream) {
+  long total_threads = (long)B * H * L * 16;
+  int grid = (int)((total_threads + 255) / 256);
+  fa_dot_packed_kernel<<<grid, 256, 0, stream>>>(
+      (const short*)dout, (const short*)o, (float*)ddot, B, H, L);
+  return hipGetLastError();
 }
 
 extern "C" hipError_t fa_dot_launch(const void* dout, const void* o,

@@ -44,14 +44,26 @@ hipError_t flash_bwd_fused_launch(const void*, const void*, const void*,
                                   const void*, void*, void*, void*, int, int,
                                   int, float, hipStream_t);
 hipError_t fa_dot_launch(const void*, const void*, void*, long, hipStream_t);
+hipError_t fa_dot_packed_launch(const void*, const void*, void*, int, int,
+                                int, hipStream_t);
+hipError_t flash_fwd_packed_launch(const void*, const void*, void*, void*,
+                                   int, int, int, float, hipStream_t);
+hipError_t flash_bwd_fused_packed_launch(const void*, const void*,
+                                         const void*, const void*,
+                                         const void*, void*, int, int, int,
+                                         float, hipStream_t);
+hipError_t flash_dq_recompute_packed_launch(const void*, const void*,
+                                            const void*, const void*,
+                                            const void*, void*, int, int,
+                                            int, float, hipStream_t);
 hipError_t flash_dq_recompute_launch(const void*, const void*, const void*,
                                      const void*, const void*, const void*,
                                      const void*, void*, int, int, int, float,
                                      hipStream_t);
 hipError_t flash_dq_launch(const void*, const void*, void*, int, int, int,
                            hipStream_t);
-hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*, int,
-                                  int, int, hipStream_t);
+hipError_t masked_pool_fwd_launch(const void*, const void*, void*, void*,
+                                  void*, int, int, int, hipStream_t);
 hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
                                   void*, int, int, int, hipStream_t);
 hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
@@ -416,6 +428,56 @@ torch::Tensor bias_grad(torch::Tensor dy) {
   return out;
 }
 
+std::vector<torch::Tensor> flash_fwd_packed(torch::Tensor qkv, long H,
+                                            c10::optional<torch::Tensor> mask,
+                                            double scale) {
+  // qkv: [B, L, 3D] packed projection output, D = H*64
+  check_bf16(qkv, "qkv");
+  TORCH_CHECK(qkv.dim() == 3, "qkv must be [B, L, 3D]");
+  const long B = qkv.size(0), L = qkv.size(1), D3 = qkv.size(2);
+  TORCH_CHECK(D3 == 3 * H * 64, "qkv last dim must be 3*H*64");
+  TORCH_CHECK(L % 32 == 0, "L % 32 == 0 required");
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    mptr = mask->data_ptr();
+  }
+  auto o = torch::empty({B, L, D3 / 3}, qkv.options());
+  auto lse = torch::empty({B, H, L}, qkv.options().dtype(torch::kFloat32));
+  CHECK_HIP(flash_fwd_packed_launch(qkv.data_ptr(), mptr, o.data_ptr(),
+                                    lse.data_ptr(), (int)B, (int)H, (int)L,
+                                    (float)scale, cur_stream()));
+  return {o, lse};
+}
+
+torch::Tensor flash_bwd_packed(torch::Tensor qkv, torch::Tensor o,
+                               torch::Tensor dout,
+                               c10::optional<torch::Tensor> mask,
+                               torch::Tensor lse, long H, double scale) {
+  // full packed backward: ddot + dK/dV + dQ, all into one [B, L, 3D] grad
+  check_bf16(qkv, "qkv"); check_bf16(o, "o"); check_bf16(dout, "dout");
+  check_f32(lse, "lse");
+  const long B = qkv.size(0), L = qkv.size(1), D3 = qkv.size(2);
+  TORCH_CHECK(D3 == 3 * H * 64 && L % 32 == 0, "bad packed shapes");
+  const void* mptr = nullptr;
+  if (mask.has_value()) {
+    check_f32(*mask, "mask");
+    mptr = mask->data_ptr();
+  }
+  auto ddot = torch::empty({B, H, L}, qkv.options().dtype(torch::kFloat32));
+  CHECK_HIP(fa_dot_packed_launch(dout.data_ptr(), o.data_ptr(),
+                                 ddot.data_ptr(), (int)B, (int)H, (int)L,
+                                 cur_stream()));
+  auto dqkv = torch::empty_like(qkv);
+  CHECK_HIP(flash_bwd_fused_packed_launch(
+      qkv.data_ptr(), dout.data_ptr(), mptr, lse.data_ptr(), ddot.data_ptr(),
+      dqkv.data_ptr(), (int)B, (int)H, (int)L, (float)scale, cur_stream()));
+  CHECK_HIP(flash_dq_recompute_packed_launch(
+      qkv.data_ptr(), dout.data_ptr(), mptr, lse.data_ptr(), ddot.data_ptr(),
+      dqkv.data_ptr(), (int)B, (int)H, (int)L, (float)scale, cur_stream()));
+  return dqkv;
+}
+
 torch::Tensor flash_dq_recompute(torch::Tensor q, torch::Tensor k,
                                  torch::Tensor v, torch::Tensor dout,
                                  c10::optional<torch::Tensor> mask,
@@ -464,9 +526,10 @@ std::vector<torch::Tensor> masked_pool_fwd(torch::Tensor x,
   }
   auto pooled = torch::empty({B, D}, x.options());
   auto counts = torch::empty({B}, x.options().dtype(torch::kFloat32));
+  auto ws = torch::empty({B, 16, D}, x.options().dtype(torch::kFloat32));
   CHECK_HIP(masked_pool_fwd_launch(x.data_ptr(), mptr, pooled.data_ptr(),
-                                   counts.data_ptr(), (int)B, (int)L, (int)D,
-                                   cur_stream()));
+                                   counts.data_ptr(), ws.data_ptr(), (int)B,
+                                   (int)L, (int)D, cur_stream()));
   return {pooled, counts};
 }
 
@@ -505,6 +568,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("emit_ds") = false);
   m.def("flash_dq_recompute", &flash_dq_recompute,
         "dQ by recompute: S/dP/dS in-register, no dS materialization");
+  m.def("flash_fwd_packed", &flash_fwd_packed,
+        "flash fwd on the packed [B,L,3D] QKV projection output");
+  m.def("flash_bwd_packed", &flash_bwd_packed,
+        "full packed flash bwd: ddot + dK/dV + dQ into one [B,L,3D] grad");
   m.def("fa_dot", &fa_dot, "rowsum(dO*O) per attention row");
   m.def("flash_dq", &flash_dq, "dQ = dS @ K (MFMA, tr_b16 K^T fragments)");
   m.def("bias_grad", &bias_grad, "bf16 column-sum for linear bias grads");

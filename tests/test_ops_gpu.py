@@ -470,3 +470,80 @@ def test_kernels_bitwise_deterministic():
     d2 = ext.bias_gelu_bwd(bgd, bgx, bgb)
     for a, b in zip(d1, d2):
         assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_flash_packed_vs_unpacked():
+    """Packed-layout flash (straight on [B, L, 3D] QKV projection output)
+    must match the unpacked kernels bit-for-bit in forward and backward —
+    same compute, different addressing geometry."""
+    torch.manual_seed(27)
+    B, H, L = 2, 4, 256
+    D = H * 64
+    qkv = _bf16(torch.randn(B, L, 3 * D))
+    mask = torch.zeros(B, L)
+    mask[:, 200:] = -1e9
+    mg = mask.cuda().contiguous()
+    scale = 0.125
+    ext = ops.hip_ops()
+    # unpacked reference path
+    q, k, v = (t.view(B, L, H, 64).transpose(1, 2).contiguous()
+               for t in qkv.split(D, dim=-1))
+    o_u, lse_u = ext.flash_fwd(q, k, v, mg, scale)
+    o_p, lse_p = ext.flash_fwd_packed(qkv, H, mg, scale)
+    assert torch.equal(lse_p, lse_u.view(B, H, L))
+    o_p4 = o_p.view(B, L, H, 64).transpose(1, 2).contiguous()
+    assert torch.equal(o_p4, o_u)
+    # backward parity
+    do = _bf16(torch.randn(B, L, D))
+    do4 = do.view(B, L, H, 64).transpose(1, 2).contiguous()
+    ddot = ext.fa_dot(do4, o_u)
+    dk_u, dv_u = ext.flash_bwd_fused(q, k, v, do4, mg, lse_u, ddot, scale)
+    dq_u = ext.flash_dq_recompute(q, k, v, do4, mg, lse_u, ddot, scale)
+    dqkv = ext.flash_bwd_packed(qkv, o_p, do, mg, lse_p, H, scale)
+    dq_p, dk_p, dv_p = (t.view(B, L, H, 64).transpose(1, 2).contiguous()
+                        for t in dqkv.split(D, dim=-1))
+    assert torch.equal(dq_p, dq_u)
+    assert torch.equal(dk_p, dk_u)
+    assert torch.equal(dv_p, dv_u)
+
+
+@pytest.mark.gpu
+def test_flash_packed_autograd_vs_cpu():
+    """End-to-end packed autograd vs the fp32 CPU fallback of the same
+    Function (bf16-sized tolerances)."""
+    torch.manual_seed(28)
+    B, H, L = 2, 2, 128
+    D = H * 64
+    qkv_cpu = torch.randn(B, L, 3 * D).to(torch.bfloat16).requires_grad_()
+    mask = torch.zeros(B, L)
+    mask[:, 100:] = -1e9
+    out_cpu = ops.flash_attention_packed(qkv_cpu, H, mask, 0.125)
+    gout = torch.randn(B, L, D).to(torch.bfloat16)
+    out_cpu.backward(gout)
+    qkv_gpu = qkv_cpu.detach().clone().cuda().requires_grad_()
+    out_gpu = ops.flash_attention_packed(qkv_gpu, H,
+                                         mask.cuda().contiguous(), 0.125)
+    out_gpu.backward(gout.cuda())
+    assert torch.allclose(out_gpu.float().cpu(), out_cpu.float(),
+                          atol=3e-2, rtol=3e-2)
+    assert torch.allclose(qkv_gpu.grad.float().cpu(), qkv_cpu.grad.float(),
+                          atol=8e-2, rtol=5e-2), \
+        (qkv_gpu.grad.float().cpu() - qkv_cpu.grad.float()).abs().max()
+
+
+@pytest.mark.gpu
+def test_flash_packed_deterministic():
+    torch.manual_seed(29)
+    B, H, L = 2, 4, 512
+    D = H * 64
+    qkv = _bf16(torch.randn(B, L, 3 * D))
+    do = _bf16(torch.randn(B, L, D))
+    mask = torch.zeros(B, L).cuda().contiguous()
+    ext = ops.hip_ops()
+    o1, l1 = ext.flash_fwd_packed(qkv, H, mask, 0.125)
+    o2, l2 = ext.flash_fwd_packed(qkv, H, mask, 0.125)
+    assert torch.equal(o1, o2) and torch.equal(l1, l2)
+    g1 = ext.flash_bwd_packed(qkv, o1, do, mask, l1, H, 0.125)
+    g2 = ext.flash_bwd_packed(qkv, o1, do, mask, l1, H, 0.125)
+    assert torch.equal(g1, g2)

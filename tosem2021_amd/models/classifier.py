@@ -99,16 +99,20 @@ class Attention(nn.Module):
 
     def forward(self, x, mask_bias: Optional[torch.Tensor]):
         B, L, D = x.shape
-        # single gather kernel to bmm-ready [3, B, H, L, dh] (csrc/repack.hip)
-        # instead of torch permute-copies + CatArrayBatchedCopy in backward
-        q, k, v = ops.qkv_repack(self.qkv(x), self.n_heads)  # [B, H, L, dh]
         if ops.flash_supported(self.head_dim, L):
-            # MFMA flash kernel: no [L, L] score tensor, O(L) memory
-            out = ops.flash_attention(q, k, v, mask_bias, self.scale)
-        else:
-            scores = torch.matmul(q, k.transpose(-1, -2))  # [B, H, L, L]
-            p = ops.fused_softmax(scores, mask_bias, self.scale)
-            out = torch.matmul(p, v)                       # [B, H, L, dh]
+            # MFMA flash kernels run straight on the packed [B, L, 3D]
+            # projection output and write attention out (and, in backward,
+            # every gradient) in the packed layouts — no repack kernels,
+            # no [L, L] score tensor, O(L) attention memory
+            out = ops.flash_attention_packed(self.qkv(x), self.n_heads,
+                                             mask_bias, self.scale)
+            return self.proj(out)
+        # general-shape fallback: gather kernel to bmm-ready
+        # [3, B, H, L, dh] (csrc/repack.hip), bmm + fused softmax
+        q, k, v = ops.qkv_repack(self.qkv(x), self.n_heads)  # [B, H, L, dh]
+        scores = torch.matmul(q, k.transpose(-1, -2))  # [B, H, L, L]
+        p = ops.fused_softmax(scores, mask_bias, self.scale)
+        out = torch.matmul(p, v)                       # [B, H, L, dh]
         return self.proj(ops.out_repack(out))
 
 

@@ -304,6 +304,64 @@ def flash_attention(q, k, v, mask=None, scale: float = 1.0):
                                    v.contiguous(), mask, scale)
 
 
+class _FlashAttentionPackedFn(torch.autograd.Function):
+    """Flash attention straight on the packed QKV projection output
+    [B, L, 3D] -> attention output [B, L, D] (round 2).  The strided-
+    geometry kernels read q/k/v from and write every gradient back into
+    the packed layouts, eliminating the qkv_repack / out_repack kernel
+    quartet (fwd gather, bwd3 merge, out fwd/bwd) from the step."""
+
+    @staticmethod
+    def forward(ctx, qkv, n_heads, mask, scale):
+        if qkv.is_cuda:
+            o, lse = hip_ops().flash_fwd_packed(qkv, n_heads, mask, scale)
+        else:
+            B, L, D3 = qkv.shape
+            d = D3 // 3
+            q, k, v = (t.view(B, L, n_heads, 64).transpose(1, 2).contiguous()
+                       for t in qkv.split(d, dim=-1))
+            o4, lse = reference.flash_attention_fwd(q, k, v, mask, scale)
+            o = o4.transpose(1, 2).reshape(B, L, d)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.n_heads = n_heads
+        ctx.mask = mask
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        n_heads, mask, scale = ctx.n_heads, ctx.mask, ctx.scale
+        do = do.contiguous()
+        if qkv.is_cuda:
+            dqkv = hip_ops().flash_bwd_packed(qkv, o, do, mask, lse,
+                                              n_heads, scale)
+            return dqkv, None, None, None
+        B, L, D3 = qkv.shape
+        d = D3 // 3
+        q, k, v = (t.view(B, L, n_heads, 64).transpose(1, 2).contiguous()
+                   for t in qkv.split(d, dim=-1))
+        do4 = do.view(B, L, n_heads, 64).transpose(1, 2).contiguous()
+        s = torch.matmul(q, k.transpose(-1, -2))
+        p = reference.p_from_lse(s, mask, lse, scale)
+        dp = torch.matmul(do4, v.transpose(-1, -2))
+        dsc = reference.softmax_bwd(dp, p, scale)
+        dq = torch.matmul(dsc, k)
+        dk = torch.matmul(dsc.transpose(-1, -2), q)
+        dv = torch.matmul(p.transpose(-1, -2), do4)
+        dqkv = torch.cat(
+            [t.transpose(1, 2).reshape(B, L, d) for t in (dq, dk, dv)],
+            dim=-1)
+        return dqkv, None, None, None
+
+
+def flash_attention_packed(qkv, n_heads: int, mask=None, scale: float = 1.0):
+    """qkv: [B, L, 3*n_heads*64] bf16 (the fused projection output);
+    returns [B, L, n_heads*64] ready for the output projection."""
+    return _FlashAttentionPackedFn.apply(qkv.contiguous(), n_heads, mask,
+                                         scale)
+
+
 def flash_supported(head_dim: int, L: int) -> bool:
     return head_dim == 64 and L % 32 == 0
 
