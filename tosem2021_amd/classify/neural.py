@@ -159,7 +159,9 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                      eval_every: int = 0, seed: int = 0,
                      dropout: float = 0.0,
                      pretrain_path: Optional[str] = None,
-                     pretrain_steps: int = 0) -> dict:
+                     pretrain_steps: int = 0,
+                     focal_gamma_property: float = 0.0,
+                     label_smoothing: float = 0.0) -> dict:
     """Fine-tune MLTC on `taxonomy_path`'s labeled rows.
 
     With `pretrain_path`/`pretrain_steps`, first train on that (typically
@@ -183,6 +185,10 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
     trainer = Trainer(tcfg, device=dev, model_cfg=cfg)
     trainer.model.set_pos_weights(
         {k: v.to(dev) for k, v in train_ds.pos_weights().items()})
+    if focal_gamma_property or label_smoothing:
+        trainer.model.set_loss_options(
+            focal_gamma={"property": focal_gamma_property},
+            label_smoothing=label_smoothing)
     if resume and ckpt_dir:
         trainer.load_or_init()
 

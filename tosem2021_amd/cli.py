@@ -137,7 +137,9 @@ def cmd_train(args):
         taxonomy_path=args.taxonomy, model=args.model, steps=args.steps,
         batch=args.batch, seq=args.seq, lr=args.lr, ckpt_dir=args.ckpt_dir,
         resume=args.resume, dropout=args.dropout,
-        pretrain_path=args.pretrain, pretrain_steps=args.pretrain_steps)
+        pretrain_path=args.pretrain, pretrain_steps=args.pretrain_steps,
+        focal_gamma_property=args.focal_gamma_property,
+        label_smoothing=args.label_smoothing, eval_every=args.eval_every)
     print(json.dumps(res, indent=2))
 
 
@@ -220,6 +222,9 @@ def main(argv=None):
     p.add_argument("--pretrain", default=None,
                    help="mined taxonomy to pretrain on before fine-tuning")
     p.add_argument("--pretrain-steps", type=int, default=0)
+    p.add_argument("--focal-gamma-property", type=float, default=0.0)
+    p.add_argument("--label-smoothing", type=float, default=0.0)
+    p.add_argument("--eval-every", type=int, default=0)
     p.set_defaults(fn=cmd_train)
 
     args = ap.parse_args(argv)

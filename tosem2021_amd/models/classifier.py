@@ -221,19 +221,42 @@ class MLTC(nn.Module):
         for name, w in weights.items():
             self.register_buffer(f"_pw_{name}", w.float(), persistent=False)
 
+    def set_loss_options(self, focal_gamma: Dict[str, float] = None,
+                         label_smoothing: float = 0.0) -> None:
+        """Per-head focal-BCE gamma (multilabel heads) + label smoothing —
+        round-2 levers for the rare-class property head.
+        """
+        self._focal_gamma = dict(focal_gamma or {})
+        self._label_smoothing = float(label_smoothing)
+
     def loss(self, logits: Dict[str, torch.Tensor],
              labels: Dict[str, torch.Tensor]) -> torch.Tensor:
         total = None
+        gammas = getattr(self, "_focal_gamma", {})
+        smooth = getattr(self, "_label_smoothing", 0.0)
         for name, lg in logits.items():
             if name not in labels:
                 continue
             lg = lg.float()
             if name in MULTILABEL_HEADS:
                 pw = getattr(self, f"_pw_{name}", None)
-                li = F.binary_cross_entropy_with_logits(
-                    lg, labels[name].float(), pos_weight=pw)
+                y = labels[name].float()
+                if smooth > 0:
+                    y = y * (1.0 - smooth) + 0.5 * smooth
+                gamma = gammas.get(name, 0.0)
+                if gamma > 0:
+                    # focal BCE: down-weight easy negatives/positives
+                    bce = F.binary_cross_entropy_with_logits(
+                        lg, y, pos_weight=pw, reduction="none")
+                    p = torch.sigmoid(lg)
+                    p_t = p * y + (1 - p) * (1 - y)
+                    li = ((1 - p_t).clamp(min=1e-4) ** gamma * bce).mean()
+                else:
+                    li = F.binary_cross_entropy_with_logits(
+                        lg, y, pos_weight=pw)
             else:
-                li = F.cross_entropy(lg, labels[name])
+                li = F.cross_entropy(lg, labels[name],
+                                     label_smoothing=smooth)
             total = li if total is None else total + li
         assert total is not None, "no labels matched any head"
         return total
