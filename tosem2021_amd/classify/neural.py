@@ -161,7 +161,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                      pretrain_path: Optional[str] = None,
                      pretrain_steps: int = 0,
                      focal_gamma_property: float = 0.0,
-                     label_smoothing: float = 0.0) -> dict:
+                     label_smoothing: float = 0.0,
+                     dump_probs_path: Optional[str] = None) -> dict:
     """Fine-tune MLTC on `taxonomy_path`'s labeled rows.
 
     With `pretrain_path`/`pretrain_steps`, first train on that (typically
@@ -222,6 +223,14 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
     ev_tuned = evaluate(trainer, val_ds, tok, seq, thresholds=ths)
     ev["strategy_micro_f1_tuned"] = round(ev_tuned["strategy_micro_f1"], 4)
     ev["property_micro_f1_tuned"] = round(ev_tuned["property_micro_f1"], 4)
+    if dump_probs_path:
+        # sigmoid probs + gold for the val AND train splits — enables
+        # offline ensembling/blending without the GPU
+        vp, vg = _head_probs(trainer, val_ds, tok, seq)
+        tp_, tg = _head_probs(trainer, train_ds, tok, seq)
+        torch.save({"val_probs": vp, "val_gold": vg,
+                    "train_probs": tp_, "train_gold": tg,
+                    "seed": seed}, dump_probs_path)
     if ckpt_dir:
         trainer.save()
     return {

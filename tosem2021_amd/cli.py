@@ -139,7 +139,8 @@ def cmd_train(args):
         resume=args.resume, dropout=args.dropout,
         pretrain_path=args.pretrain, pretrain_steps=args.pretrain_steps,
         focal_gamma_property=args.focal_gamma_property,
-        label_smoothing=args.label_smoothing, eval_every=args.eval_every)
+        label_smoothing=args.label_smoothing, eval_every=args.eval_every,
+        seed=args.seed, dump_probs_path=args.dump_probs)
     print(json.dumps(res, indent=2))
 
 
@@ -225,6 +226,10 @@ def main(argv=None):
     p.add_argument("--focal-gamma-property", type=float, default=0.0)
     p.add_argument("--label-smoothing", type=float, default=0.0)
     p.add_argument("--eval-every", type=int, default=0)
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--dump-probs", default=None,
+                   help="save val/train sigmoid probs + gold for offline "
+                        "ensembling")
     p.set_defaults(fn=cmd_train)
 
     args = ap.parse_args(argv)
