@@ -1,19 +1,26 @@
 #!/usr/bin/env python3
-"""Prove RCCL executes on MI355X.
+"""Prove RCCL executes on MI355X within a 1-GPU lease.
 
-Two ranks share one GPU (RCCL supports multi-rank-per-device), nccl(=RCCL)
-backend:
-  1. allreduce correctness: sum of per-rank constants;
-  2. allreduce timing sweep (1 MiB .. 256 MiB bf16) — same-device, so the
-     numbers measure RCCL's kernel path, not xGMI wires;
-  3. three DDP training steps of the mltc model through parallel/ddp.py's
-     bucketed async allreduce on the nccl backend, losses compared across
-     ranks (they see identical data -> identical loss).
+RCCL 2.26 (like NCCL >= 2.5) REFUSES two ranks on one device:
+    ncclInvalidUsage ... Duplicate GPU detected: rank 1 and rank 0 both on
+    CUDA device 8e000
+(measured on-box, see profiles/rccl_proof.md) — so the round-1 VERDICT's
+suggestion of a 2-rank/1-GPU bench is impossible on this stack.  What CAN
+be proven on one GPU, and is proven here:
 
-Launch (single MI355X box):
-    python -m torch.distributed.run --nnodes 1 --nproc-per-node 2 \
-        --master-addr 127.0.0.1 --standalone scripts/rccl_proof.py
-Writes rank-0 results JSON to gpurun_out/rccl_proof.json (or stdout).
+  1. the nccl(=RCCL) backend initializes a communicator on the MI355X
+     (librccl.so version banner + init);
+  2. RCCL collectives EXECUTE: world-1 all_reduce / broadcast /
+     all_gather_into_tensor / reduce_scatter_tensor enqueue RCCL kernels
+     (run with NCCL_DEBUG=INFO to capture the collective launch records);
+  3. correctness + a timing sweep over payload sizes.
+
+The multi-device path itself is exercised by the driver's round-end 8-GPU
+scaling bench; the bucketed-allreduce DDP logic is covered by the
+world_size 2/4 gloo tests (tests/test_ddp_cpu.py).
+
+Launch:  python scripts/rccl_proof.py        (single process, 1 GPU)
+Writes gpurun_out/rccl_proof.json.
 """
 import json
 import os
@@ -28,24 +35,34 @@ import torch.distributed as dist
 
 
 def main() -> None:
-    rank = int(os.environ["RANK"])
-    world = int(os.environ["WORLD_SIZE"])
-    # both ranks pin the same physical GPU
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29513")
     torch.cuda.set_device(0)
-    dist.init_process_group("nccl", rank=rank, world_size=world)
+    dist.init_process_group("nccl", rank=0, world_size=1)
     dev = torch.device("cuda", 0)
-    res = {"backend": dist.get_backend(), "world_size": world,
+    res = {"backend": dist.get_backend(), "world_size": 1,
            "nccl_version": list(torch.cuda.nccl.version()),
-           "device": torch.cuda.get_device_name(0)}
+           "device": torch.cuda.get_device_name(0),
+           "duplicate_gpu_refusal":
+               "RCCL 2.26 refuses >1 rank per device (ncclInvalidUsage: "
+               "Duplicate GPU detected) — measured 2026-09-14, full "
+               "traceback in profiles/rccl_proof.md"}
 
-    # 1. correctness
-    t = torch.full((1024,), float(rank + 1), device=dev)
+    # collectives execute + are correct
+    t = torch.full((1 << 20,), 3.0, device=dev)
     dist.all_reduce(t)
-    expect = sum(range(1, world + 1))
-    assert torch.all(t == expect), t[:4]
-    res["allreduce_correct"] = True
+    assert torch.all(t == 3.0)
+    dist.broadcast(t, src=0)
+    out = torch.empty_like(t)
+    dist.all_gather_into_tensor(out, t)
+    assert torch.all(out == 3.0)
+    rs = torch.empty_like(t)
+    dist.reduce_scatter_tensor(rs, t)
+    assert torch.all(rs == 3.0)
+    res["collectives_correct"] = ["all_reduce", "broadcast",
+                                  "all_gather_into_tensor",
+                                  "reduce_scatter_tensor"]
 
-    # 2. timing sweep (bf16, sizes in MiB)
     sweep = {}
     for mib in (1, 16, 64, 256):
         n = mib * (1 << 20) // 2
@@ -53,49 +70,19 @@ def main() -> None:
         for _ in range(3):
             dist.all_reduce(x)
         torch.cuda.synchronize()
-        dist.barrier()
         t0 = time.perf_counter()
         iters = 10
         for _ in range(iters):
             dist.all_reduce(x)
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / iters
-        # algbw convention: bytes / time
-        sweep[f"{mib}MiB"] = {"us": round(dt * 1e6, 1),
-                              "algbw_GBps": round(mib / 1024 / dt, 1)}
-    res["allreduce_sweep"] = sweep
+        sweep[f"{mib}MiB"] = {"us": round(dt * 1e6, 1)}
+    res["allreduce_sweep_world1"] = sweep
 
-    # 3. DDP training steps through parallel/ddp.py on RCCL
-    from tosem2021_amd.data.synthetic import synthetic_batch
-    from tosem2021_amd.models.classifier import CONFIGS
-    from tosem2021_amd.train import TrainConfig, Trainer
-
-    torch.manual_seed(7)
-    trainer = Trainer(TrainConfig(model="mltc-base", warmup_steps=0),
-                      device=dev)
-    assert trainer.ddp.enabled and trainer.ddp.world_size == world
-    cfg = CONFIGS["mltc-base"]
-    losses = []
-    for _ in range(3):
-        tokens, mask, labels = synthetic_batch(cfg, 8, 256, device=dev,
-                                               seed=11)
-        losses.append(trainer.step(tokens, mask, labels))
-    # identical data on both ranks -> allreduced grads == local grads,
-    # losses must match across ranks
-    lt = torch.tensor(losses, device=dev)
-    gathered = [torch.empty_like(lt) for _ in range(world)]
-    dist.all_gather(gathered, lt)
-    max_dev = max(float((g - gathered[0]).abs().max()) for g in gathered)
-    res["ddp_steps_losses"] = [round(x, 4) for x in losses]
-    res["cross_rank_loss_max_dev"] = max_dev
-    assert max_dev < 1e-6, max_dev
-    res["ddp_buckets"] = len(trainer.ddp._buckets)
-
-    if rank == 0:
-        os.makedirs("gpurun_out", exist_ok=True)
-        with open("gpurun_out/rccl_proof.json", "w") as f:
-            json.dump(res, f, indent=1)
-        print(json.dumps(res))
+    os.makedirs("gpurun_out", exist_ok=True)
+    with open("gpurun_out/rccl_proof.json", "w") as f:
+        json.dump(res, f, indent=1)
+    print(json.dumps(res))
     dist.destroy_process_group()
 
 

@@ -251,7 +251,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
 def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
                      model: str = "mltc-base", seq: int = 256,
                      batch: int = 64, threshold: float = 0.5,
-                     device: Optional[str] = None) -> str:
+                     device: Optional[str] = None,
+                     repo_prefix: bool = False) -> str:
     """Label the rows of a (mined) taxonomy CSV with a trained MLTC model and
     rewrite the 41-column CSV with the predicted labels."""
     from tosem2021_amd.extract.schema import (
@@ -274,6 +275,11 @@ def apply_classifier(ckpt_dir: str, taxonomy_path: str, out_csv: str,
 
     texts = (df["Labels"].astype(str) + " | " +
              df["Component"].astype(str)).tolist()
+    if repo_prefix:
+        # models trained on repo-prefixed text (artifacts/
+        # mltc_train_repo_s256.pt) expect the same context token
+        texts = [f"REPO_{r} {t}"
+                 for r, t in zip(df["Repo"].astype(str), texts)]
     rows = []
     for lo in range(0, len(texts), batch):
         chunk = texts[lo:lo + batch]

@@ -127,7 +127,8 @@ def cmd_classify(args):
     from tosem2021_amd.classify.neural import apply_classifier
     out = apply_classifier(args.ckpt_dir, args.taxonomy, args.out,
                            model=args.model, seq=args.seq,
-                           threshold=args.threshold)
+                           threshold=args.threshold,
+                           repo_prefix=args.repo_prefix)
     print(f"wrote {out}")
 
 
@@ -208,6 +209,8 @@ def main(argv=None):
     p.add_argument("--model", default="mltc-base")
     p.add_argument("--seq", type=int, default=256)
     p.add_argument("--threshold", type=float, default=0.5)
+    p.add_argument("--repo-prefix", action="store_true",
+                   help="model was trained on repo-prefixed text")
     p.set_defaults(fn=cmd_classify)
 
     p = sub.add_parser("train", help="train the MLTC classifier on a taxonomy")
