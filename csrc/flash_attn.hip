@@ -38,7 +38,6 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 //  alpha:  [4 waves][32] f32                                   =  512 B
 #define K_LDS_BYTES (FA_KVB * 128)
 #define VT_LDS_BYTES (FA_KVB * 128)
-#define KV2_LDS_BYTES (2 * FA_KVB * 128)   // 64-row stage (fwd, round 2)
 
 __device__ __forceinline__ int kswz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
@@ -65,9 +64,9 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // softmax chain (ILP within the wave).  2 waves/SIMD by registers; the
   // workgroup covers 256 q rows.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = (short*)smem;                       // swizzled [64][64]
-  short* v_lds = (short*)(smem + KV2_LDS_BYTES);     // swizzled [64][64]
-  float* alpha_lds = (float*)(smem + 2 * KV2_LDS_BYTES);
+  short* k_lds = (short*)smem;                       // swizzled [32][64]
+  short* v_lds = (short*)(smem + K_LDS_BYTES);       // swizzled [32][64]
+  float* alpha_lds = (float*)(smem + K_LDS_BYTES + VT_LDS_BYTES);
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -109,43 +108,26 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   for (int t = 0; t < 2; ++t) { oA[t] = (f32x16)(0.f); oB[t] = (f32x16)(0.f); }
   float mA = -3.0e38f, lA = 0.f, mB = -3.0e38f, lB = 0.f;
 
-  // 64 kv rows staged per barrier pair (round 2): each thread stages TWO
-  // 16-B packets; the two 32-row MFMA rounds then run back-to-back off the
-  // same stage, halving the per-kv-row barrier count (the kernels are
-  // barrier/latency-bound at 2 waves/SIMD — profiles/r23_flash_pmc.md).
-  const int n_st = (L + 2 * FA_KVB - 1) / (2 * FA_KVB);  // ceil: L%64==32 tail
+  const int n_kv = L / FA_KVB;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
-  short8_t kv8[2], vv8[2];
-#pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    int row = min(u * FA_KVB + srow, L - 1);   // clamp for the 32-row tail
-    kv8[u] = *(const short8_t*)(k + qkv_off + (long)row * qkv_rs + (sc8 >> 1));
-    vv8[u] = *(const short8_t*)(v + qkv_off + (long)row * qkv_rs + (sc8 >> 1));
-  }
-  for (int st = 0; st < n_st; ++st) {
+  short8_t kv8 = *(const short8_t*)(k + qkv_off + (long)srow * qkv_rs +
+                                    (sc8 >> 1));
+  short8_t vv8 = *(const short8_t*)(v + qkv_off + (long)srow * qkv_rs +
+                                    (sc8 >> 1));
+  for (int kt = 0; kt < n_kv; ++kt) {
+    const int kv0 = kt * FA_KVB;
     __syncthreads();
-#pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      int r64 = u * FA_KVB + srow;
-      *(short8_t*)((char*)k_lds + r64 * 128 + kswz(r64, sc8)) = kv8[u];
-      *(short8_t*)((char*)v_lds + r64 * 128 + kswz(r64, sc8)) = vv8[u];
+    {
+      *(short8_t*)((char*)k_lds + srow * 128 + kswz(srow, sc8)) = kv8;
+      *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
     }
     __syncthreads();
-    if (st + 1 < n_st) {
-#pragma unroll
-      for (int u = 0; u < 2; ++u) {
-        int nrow = min((st + 1) * 2 * FA_KVB + u * FA_KVB + srow, L - 1);
-        kv8[u] = *(const short8_t*)(k + qkv_off + (long)nrow * qkv_rs +
-                                    (sc8 >> 1));
-        vv8[u] = *(const short8_t*)(v + qkv_off + (long)nrow * qkv_rs +
-                                    (sc8 >> 1));
-      }
+    if (kt + 1 < n_kv) {
+      kv8 = *(const short8_t*)(k + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
+      vv8 = *(const short8_t*)(v + qkv_off +
+                               (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
     }
-#pragma unroll
-  for (int sub = 0; sub < 2; ++sub) {
-    const int kv0 = (st * 2 + sub) * FA_KVB;
-    if (kv0 >= L) break;                       // odd-32 tail: one sub-round
-    const int lds_base = sub * (FA_KVB * 128);
 
     // ---- QK^T for BOTH q-blocks (8 back-to-back MFMAs) ----
     f32x16 sA = (f32x16)(0.f), sB = (f32x16)(0.f);
@@ -153,9 +135,8 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int byte_off = (16 * c + 8 * half) * 2;
-      int lrow = sub * FA_KVB + col;
-      short8_t kf = *(const short8_t*)((char*)k_lds + lrow * 128 +
-                                       kswz(lrow, byte_off));
+      short8_t kf = *(const short8_t*)((char*)k_lds + col * 128 +
+                                       kswz(col, byte_off));
       sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfA[c], sA, 0, 0, 0);
       sB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfB[c], sB, 0, 0, 0);
     }
@@ -235,7 +216,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     // ---- PV for both blocks from tr_b16-gathered V fragments ----
     {
       typedef __attribute__((ext_vector_type(2))) unsigned uint2_t;
-      const unsigned vbase = (unsigned)(unsigned long)((char*)v_lds + lds_base);
+      const unsigned vbase = (unsigned)(unsigned long)(char*)v_lds;
       const int kv_mate = (lane >> 2) & 3;
       const int d_lane = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
       unsigned a[8];
@@ -288,7 +269,6 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       __builtin_amdgcn_s_setprio(0);
     }
   }
-  }
 
   // ---- epilogue: both blocks ----
   if (lse != nullptr && half == 0) {
@@ -320,7 +300,7 @@ extern "C" hipError_t flash_fwd_launch(const void* q, const void* k,
                                        hipStream_t stream) {
   int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_qblocks);
-  size_t shm = 2 * KV2_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const float*)mask,
       (short*)o, (float*)lse, B, H, L, scale,
@@ -338,7 +318,7 @@ extern "C" hipError_t flash_fwd_packed_launch(const void* qkv,
   const int D = H * FA_DH;
   int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_qblocks);
-  size_t shm = 2 * KV2_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + FA_WAVES * 64 * sizeof(float);
   flash_fwd_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)qkv, (const short*)qkv + D, (const short*)qkv + 2 * D,
       (const float*)mask, (short*)o, (float*)lse, B, H, L, scale,
