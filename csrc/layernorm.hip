@@ -760,23 +760,43 @@ hipError_t ln_bwd_launch(const void* dy, const void* x, const void* gamma,
   return hipGetLastError();
 }
 
-#define COLSUM_SPLITS 64
+// Adaptive split count (round 2): the fixed 64-way split left a D=1024
+// reduction at 64 blocks on a 256-CU chip (21.6 us/call in the r2 bench
+// profile); pick splits so stage 1 launches ~1024 blocks, then collapse
+// with a narrow second stage and a single-split final stage.  scratch
+// must hold COLSUM_MAX_SPLITS + COLSUM_MID rows (ops.cpp allocates it).
+#define COLSUM_MAX_SPLITS 256
+#define COLSUM_MID 16
 hipError_t colsum_launch(const void* ws, void* scratch, void* out, int R,
                          int D, hipStream_t stream) {
-  dim3 grid1((D / 4 + 255) / 256, 1);
-  if (R <= COLSUM_SPLITS) {
+  int gx = (D / 4 + 255) / 256;
+  dim3 grid1(gx, 1);
+  if (R <= COLSUM_MID) {
     colsum_stage_kernel<<<grid1, 256, 0, stream>>>(
         (const float*)ws, (float*)out, R, D, R);
     return hipGetLastError();
   }
-  int rows_per_split = (R + COLSUM_SPLITS - 1) / COLSUM_SPLITS;
-  dim3 grid2((D / 4 + 255) / 256, COLSUM_SPLITS);
-  colsum_stage_kernel<<<grid2, 256, 0, stream>>>(
+  int splits = 1024 / gx;
+  if (splits > COLSUM_MAX_SPLITS) splits = COLSUM_MAX_SPLITS;
+  if (splits > R) splits = R;
+  int rows_per_split = (R + splits - 1) / splits;
+  colsum_stage_kernel<<<dim3(gx, splits), 256, 0, stream>>>(
       (const float*)ws, (float*)scratch, R, D, rows_per_split);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return e;
+  if (splits <= COLSUM_MID) {
+    colsum_stage_kernel<<<grid1, 256, 0, stream>>>(
+        (const float*)scratch, (float*)out, splits, D, splits);
+    return hipGetLastError();
+  }
+  float* mid = (float*)scratch + (long)COLSUM_MAX_SPLITS * D;
+  int rps2 = (splits + COLSUM_MID - 1) / COLSUM_MID;
+  colsum_stage_kernel<<<dim3(gx, COLSUM_MID), 256, 0, stream>>>(
+      (const float*)scratch, mid, splits, D, rps2);
+  e = hipGetLastError();
+  if (e != hipSuccess) return e;
   colsum_stage_kernel<<<grid1, 256, 0, stream>>>(
-      (const float*)scratch, (float*)out, COLSUM_SPLITS, D, COLSUM_SPLITS);
+      mid, (float*)out, COLSUM_MID, D, COLSUM_MID);
   return hipGetLastError();
 }
 

@@ -164,7 +164,7 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                           cur_stream()));
   auto dgamma = torch::empty({D}, opts);
   auto dbeta = torch::empty({D}, opts);
-  auto scratch = torch::empty({64, D}, opts);
+  auto scratch = torch::empty({256 + 16, D}, opts);
   CHECK_HIP(colsum_launch(ws_dg.data_ptr(), scratch.data_ptr(),
                           dgamma.data_ptr(), (int)ws_rows, D, cur_stream()));
   CHECK_HIP(colsum_launch(ws_db.data_ptr(), scratch.data_ptr(),
@@ -205,7 +205,8 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                                  dx.data_ptr(), ws.data_ptr(), n, D, grid,
                                  cur_stream()));
   auto dbias = torch::empty({D}, x.options().dtype(torch::kFloat32));
-  auto scratch = torch::empty({64, D}, x.options().dtype(torch::kFloat32));
+  auto scratch = torch::empty({256 + 16, D},
+                             x.options().dtype(torch::kFloat32));
   CHECK_HIP(colsum_launch(ws.data_ptr(), scratch.data_ptr(), dbias.data_ptr(),
                           grid, D, cur_stream()));
   return {dx, dbias};
