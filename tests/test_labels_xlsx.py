@@ -118,3 +118,34 @@ def test_case_labels_sheet(reference_mounted):
     cases = load_case_labels()
     assert len(cases) > 2000
     assert any("error-handling" in c["label"].lower() for c in cases)
+
+
+def test_selection_workbooks_consistent_with_csvs(reference_mounted):
+    """The selection XLSX workbooks (unreadable in round 1) agree with
+    their CSV exports: same repo sets for v3; Repos.xlsx is the finalist
+    list; Repos_metrics_v2.xlsx is the deduplicated per-repo view of the
+    per-topic v2 CSV."""
+    import pandas as pd
+
+    from tosem2021_amd.utils.xlsx import first_sheet
+
+    base = "/root/reference/selection/Reposition"
+    v3x = first_sheet(f"{base}/Repos_metrics_v3.xlsx")
+    v3c = pd.read_csv(f"{base}/Repos_metrics_v3.csv", encoding="utf-8-sig")
+    assert len(v3x) - 1 == len(v3c) == 311
+    xlsx_repos = {r[0].strip() for r in v3x[1:] if r and r[0].strip()}
+    csv_repos = set(v3c["Repos"].astype(str).str.strip())
+    assert xlsx_repos == csv_repos
+
+    v2x = first_sheet(f"{base}/Repos_metrics_v2.xlsx")
+    v2c = pd.read_csv(f"{base}/Repos_metrics_v2.csv", encoding="utf-8-sig")
+    v2x_repos = {r[0].strip() for r in v2x[1:] if r and r[0].strip()}
+    v2c_repos = set(v2c["Repos"].astype(str).str.strip())
+    # xlsx = per-repo dedup of the per-topic CSV (157 unique + 1 extra row)
+    assert len(v2x) - 1 == 158
+    assert v2c_repos <= v2x_repos
+
+    fx = first_sheet(f"{base}/Repos.xlsx")
+    fl = pd.read_csv(f"{base}/Repos_l.csv", encoding="utf-8-sig")
+    fx_repos = {r[0].strip() for r in fx[1:] if r and r[0].strip()}
+    assert set(fl["Repos"].astype(str).str.strip()) <= fx_repos
