@@ -16,6 +16,7 @@ through hipBLASLt (torch.matmul / F.linear on ROCm), and the non-GEMM hot ops
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass, field
 from typing import Dict, Optional
 
@@ -82,6 +83,22 @@ class FusedLayerNorm(nn.Module):
         return ops.fused_layernorm(x, self.weight, self.bias, self.eps)
 
 
+# A/B knob (TOSEM_FUSED_LINEAR=1): route the bias-carrying projections
+# through ops.fused_linear — torch addmm forward (same epilogue GEMM) but
+# backward computes db with the colsum bias_grad kernel (~20 us) instead of
+# at::native::reduce (~58 us).  Round 1 measured the explicit dgrad/wgrad
+# matmuls dispatching worse than addmm's backward; with the committed
+# TunableOp table covering those layouts this is worth re-measuring.
+_USE_FUSED_LINEAR = os.environ.get("TOSEM_FUSED_LINEAR", "0") == "1"
+
+
+def _linear(mod: nn.Linear, x: torch.Tensor) -> torch.Tensor:
+    if _USE_FUSED_LINEAR and x.is_cuda and mod.bias is not None \
+            and torch.is_grad_enabled():
+        return ops.fused_linear(x, mod.weight, mod.bias)
+    return mod(x)
+
+
 class Attention(nn.Module):
     """Multi-head bidirectional self-attention.
 
@@ -104,9 +121,9 @@ class Attention(nn.Module):
             # projection output and write attention out (and, in backward,
             # every gradient) in the packed layouts — no repack kernels,
             # no [L, L] score tensor, O(L) attention memory
-            out = ops.flash_attention_packed(self.qkv(x), self.n_heads,
+            out = ops.flash_attention_packed(_linear(self.qkv, x), self.n_heads,
                                              mask_bias, self.scale)
-            return self.proj(out)
+            return _linear(self.proj, out)
         # general-shape fallback: gather kernel to bmm-ready
         # [3, B, H, L, dh] (csrc/repack.hip), bmm + fused softmax
         q, k, v = ops.qkv_repack(self.qkv(x), self.n_heads)  # [B, H, L, dh]
@@ -134,7 +151,7 @@ class FFN(nn.Module):
             return self.down(a.view(B, L, -1))
         h = self.up(x)
         h = ops.fused_bias_gelu(h, self.up_bias)
-        return self.down(h)
+        return _linear(self.down, h)
 
 
 class EncoderBlock(nn.Module):
