@@ -86,9 +86,12 @@ class FusedLayerNorm(nn.Module):
 # A/B knob (TOSEM_FUSED_LINEAR=1): route the bias-carrying projections
 # through ops.fused_linear — torch addmm forward (same epilogue GEMM) but
 # backward computes db with the colsum bias_grad kernel (~20 us) instead of
-# at::native::reduce (~58 us).  Round 1 measured the explicit dgrad/wgrad
-# matmuls dispatching worse than addmm's backward; with the committed
-# TunableOp table covering those layouts this is worth re-measuring.
+# at::native::reduce (~58 us).  RE-MEASURED round 2 with the committed
+# TunableOp table active: still a 13.5 ms/step REGRESSION (97.6 vs 84.1 ms
+# back-to-back on one box) — the explicit dgrad/wgrad matmul layouts are
+# not in the tuned table and dispatch worse than addmm's backward, exactly
+# as in round 1.  Default stays OFF; the ~1.4 ms of at::native::reduce
+# bias grads is the price of addmm's better GEMM dispatch.
 _USE_FUSED_LINEAR = os.environ.get("TOSEM_FUSED_LINEAR", "0") == "1"
 
 
