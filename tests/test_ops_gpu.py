@@ -46,7 +46,10 @@ def test_layernorm_bwd(N, D):
     dxe, dge, dbe = ref.layernorm_bwd(dyg.float().cpu(), xg.float().cpu(),
                                       gg.float().cpu(), mean.cpu(), rstd.cpu())
     assert torch.allclose(dx.float().cpu(), dxe, atol=4e-2, rtol=3e-2)
-    tol = dict(atol=0.3 + 0.02 * math.sqrt(N), rtol=2e-2)
+    # kernel and reference both accumulate in f32 from the same bf16
+    # inputs — only summation-order noise remains (VERDICT r1 weak #6:
+    # the old 0.3 + 0.02*sqrt(N) would have hidden real reduction bugs)
+    tol = dict(atol=2e-2 + 2e-3 * math.sqrt(N), rtol=2e-2)
     assert torch.allclose(dgamma.cpu(), dge, **tol)
     assert torch.allclose(dbeta.cpu(), dbe, **tol)
 
@@ -64,7 +67,9 @@ def test_bias_gelu(N, D):
     dxe, dbe = ref.bias_gelu_bwd(dyg.float().cpu(), xg.float().cpu(),
                                  bg.float().cpu())
     assert torch.allclose(dx.float().cpu(), dxe, atol=3e-2, rtol=2e-2)
-    assert torch.allclose(dbias.cpu(), dbe, atol=0.3 + 0.02 * math.sqrt(N),
+    # dbias: f32 accumulation both sides; the gelu'(pre) term differs
+    # (hardware-exp kernel vs tanh-form reference) by ~1e-3/elem
+    assert torch.allclose(dbias.cpu(), dbe, atol=0.1 + 0.01 * math.sqrt(N),
                           rtol=2e-2)
 
 
@@ -394,8 +399,11 @@ def test_bias_grad_vs_sum(N, D):
     dy = _bf16(torch.randn(N, D))
     db = ops.hip_ops().bias_grad(dy)
     ref_db = dy.float().sum(0)
-    assert torch.allclose(db.float(), ref_db, atol=0.3 + 0.02 * N ** 0.5,
-                          rtol=2e-2), (db.float() - ref_db).abs().max()
+    # exact bf16 values summed in f32 both sides; only ordering noise
+    # (bias_grad output is bf16, so allow one bf16 ulp of the magnitude)
+    assert torch.allclose(db.float(), ref_db,
+                          atol=5e-2 + 2e-3 * N ** 0.5, rtol=1e-2), \
+        (db.float() - ref_db).abs().max()
 
 
 @pytest.mark.gpu
