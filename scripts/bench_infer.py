@@ -21,6 +21,8 @@ def main():
     ap.add_argument("--seq", type=int, default=256)
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--graph", action="store_true",
+                    help="replay the forward as one captured hipGraph")
     args = ap.parse_args()
 
     torch.manual_seed(0)
@@ -30,13 +32,19 @@ def main():
     mask = torch.ones(args.batch, args.seq, device="cuda",
                       dtype=torch.bool)
 
+    if args.graph:
+        from tosem2021_amd.utils.hipgraph import CapturedForward
+        fwd = CapturedForward(model, args.batch, args.seq)
+        run = lambda: fwd(toks, mask)  # noqa: E731
+    else:
+        run = lambda: model(toks, mask)  # noqa: E731
     with torch.no_grad():
         for _ in range(args.warmup):
-            logits = model(toks, mask)
+            logits = run()
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(args.iters):
-            logits = model(toks, mask)
+            logits = run()
             probs = {h: torch.sigmoid(v.float()) for h, v in logits.items()}
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / args.iters
@@ -49,7 +57,7 @@ def main():
         "dtype": "bf16",
         "data": "synthetic",
         "config": {"model": args.model, "batch": args.batch,
-                   "seq_len": args.seq},
+                   "seq_len": args.seq, "hipgraph": bool(args.graph)},
     }))
 
 

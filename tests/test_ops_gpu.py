@@ -555,3 +555,30 @@ def test_flash_packed_deterministic():
     g1 = ext.flash_bwd_packed(qkv, o1, do, mask, l1, H, 0.125)
     g2 = ext.flash_bwd_packed(qkv, o1, do, mask, l1, H, 0.125)
     assert torch.equal(g1, g2)
+
+
+@pytest.mark.gpu
+def test_hipgraph_captured_forward_matches_eager():
+    """hipGraph-captured inference (utils/hipgraph.py) must match the eager
+    forward on the same inputs."""
+    from tosem2021_amd.models.classifier import CONFIGS, build_model
+    from tosem2021_amd.utils.hipgraph import CapturedForward
+    torch.manual_seed(31)
+    model = build_model("mltc-tiny").cuda().eval()
+    B, L = 4, 64
+    fwd = CapturedForward(model, B, L)
+    toks = torch.randint(4, CONFIGS["mltc-tiny"].vocab_size, (B, L),
+                         device="cuda")
+    mask = torch.ones(B, L, dtype=torch.bool, device="cuda")
+    mask[:, 50:] = False
+    out_g = {k: v.clone() for k, v in fwd(toks, mask).items()}
+    with torch.no_grad():
+        out_e = model(toks, mask)
+    for k in out_e:
+        assert torch.allclose(out_g[k].float(), out_e[k].float(),
+                              atol=3e-2, rtol=3e-2), k
+    # replay with different inputs gives different outputs (buffers rebind)
+    toks2 = torch.randint(4, CONFIGS["mltc-tiny"].vocab_size, (B, L),
+                          device="cuda")
+    out_g2 = fwd(toks2, mask)
+    assert not torch.equal(out_g2["strategy"], out_g["strategy"])
