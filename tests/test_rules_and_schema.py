@@ -146,3 +146,46 @@ def test_property_lexicon_predict_shape():
         lex.save(f.name)
         lex2 = PropertyLexicon.load(f.name)
         assert lex2.predict({"T:accuracy"}) == ["Correctness"]
+
+
+def test_strategy_stack_heldout_floor():
+    """Stacked strategy labeler (rules + calibrated lexicon, per-class
+    combinators): held-out micro-F1 >= 0.68 (rules alone: 0.600)."""
+    import os
+
+    import pytest
+
+    if not os.path.exists("/root/reference/RQs/taxonomy_test2.csv"):
+        pytest.skip("reference corpus not mounted on this box")
+    from tosem2021_amd.analyze.taxonomy import load_taxonomy
+    from tosem2021_amd.classify.agreement import evaluate_rules_on_taxonomy
+    from tosem2021_amd.classify.strategy_stack import default_stack
+    assert default_stack() is not None, "artifacts/strategy_stack.json missing"
+    df = load_taxonomy("/root/reference/RQs/taxonomy_test2.csv")
+    res = evaluate_rules_on_taxonomy(df)
+    assert res["strategy_labeler"] == "stacked"
+    assert res["strategy_micro_f1_heldout"] >= 0.68, res
+
+
+def test_strategy_stack_roundtrip_row_encoding(tmp_path):
+    """apply_to_row re-encodes the stacked strategy set into the taxonomy
+    columns such that row.strategies() reads back exactly that set."""
+    from tosem2021_amd.classify import strategy_stack
+    from tosem2021_amd.classify.property_lexicon import property_features
+    from tosem2021_amd.classify.rules import classify_text
+    stack = strategy_stack.default_stack()
+    if stack is None:
+        import pytest
+        pytest.skip("strategy_stack artifact missing")
+    for text in ["assertRaises(ValueError, f)", "assertTrue(x.ok())",
+                 "assertAlmostEqual(a, b, places=3)",
+                 "EXPECT_EQ(1, count)"]:
+        row = classify_text(text)
+        feats = property_features(text, "", "auto_sklearn", row=row)
+        expect = stack.predict(feats, set(row.strategies()))
+        strategy_stack.apply_to_row(row, text, "", "auto_sklearn",
+                                    feats=feats)
+        got = set(row.strategies())
+        # error_handling preservation may keep an unmapped Error_Type but
+        # never adds a STRATEGY; the read-back must equal the prediction
+        assert got == expect, (text, got, expect)

@@ -80,13 +80,21 @@ def evaluate_rules_on_taxonomy(df: pd.DataFrame, limit: int = 0) -> dict:
     repos = df["Repo"].astype(str).tolist()
     from tosem2021_amd.classify.property_lexicon import (
         default_lexicon, property_features)
+    from tosem2021_amd.classify.strategy_stack import default_stack
     lex = default_lexicon()
+    stack = default_stack()
     for text, comp, repo in zip(texts, comps, repos):
         row = classify_text(text, name="", path=comp)
-        pred_strat.append(set(row.strategies()))
+        feats = None
+        if lex is not None or stack is not None:
+            feats = property_features(text, comp, repo, row=row)
+        rule_set = set(row.strategies())
+        if stack is not None:
+            pred_strat.append(stack.predict(feats, rule_set))
+        else:
+            pred_strat.append(rule_set)
         if lex is not None:
-            pred_props.append(set(lex.predict(
-                property_features(text, comp, repo, row=row))))
+            pred_props.append(set(lex.predict(feats)))
         else:
             pred_props.append(set(row.properties()))
         pred_method.append(row.method)
@@ -103,7 +111,13 @@ def evaluate_rules_on_taxonomy(df: pd.DataFrame, limit: int = 0) -> dict:
         "property_micro_f1": micro_f1(prop_scores),
         "method_accuracy": method_acc,
         "property_labeler": "lexicon" if lex is not None else "regex",
+        "strategy_labeler": "stacked" if stack is not None else "regex",
     }
+    if stack is not None:
+        odd = [i for i in range(len(df)) if i % 2 == 1]
+        strat_ho = _score_sets([pred_strat[i] for i in odd],
+                               [gold_strat[i] for i in odd], STRATEGIES)
+        res["strategy_micro_f1_heldout"] = micro_f1(strat_ho)
     if lex is not None:
         # the committed lexicon was fit on the even-index gold rows
         # (property_lexicon.py protocol) — report the uncontaminated

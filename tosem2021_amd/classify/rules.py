@@ -275,7 +275,9 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
     head = classify_text(
         f"{desc}: " + "; ".join(a.source for a in case.assertions[:4]),
         name=case.qualname, path=case.file_rel)
-    from tosem2021_amd.classify.property_lexicon import apply_to_row
+    from tosem2021_amd.classify import strategy_stack
+    from tosem2021_amd.classify.property_lexicon import (apply_to_row,
+                                                         property_features)
     head.repo = repo
     head.file_id = file_id
     head.component = component or case.file_rel
@@ -283,14 +285,19 @@ def classify_case(case: TestCase, repo: str, file_id: int = 0,
         * max(case.param_multiplicity, 1)
     if case.uses_mock:
         head.flags["mock_test"] = 1
+    feats = property_features(desc, head.component, repo, row=head)
     apply_to_row(head, desc, head.component, repo)
+    strategy_stack.apply_to_row(head, desc, head.component, repo, feats=feats)
     rows.append(head)
     for a in case.assertions:
         r = classify_text(a.source, name=case.qualname, path=case.file_rel)
         r.repo = repo
         r.file_id = file_id
         r.component = component or case.file_rel
+        feats = property_features(a.source, r.component, repo, row=r)
         apply_to_row(r, a.source, r.component, repo)
+        strategy_stack.apply_to_row(r, a.source, r.component, repo,
+                                    feats=feats)
         # assertion rows inherit the enclosing case's workflow stage (an
         # assertion's own text rarely carries stage cues)
         r.category = head.category
