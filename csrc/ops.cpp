@@ -69,6 +69,8 @@ hipError_t masked_pool_bwd_launch(const void*, const void*, const void*,
 hipError_t tr_probe_launch(const void*, void*, int, hipStream_t);
 hipError_t colsum_bf16_launch(const void*, void*, void*, long, int,
                               hipStream_t);
+hipError_t wgrad_gemm_launch(const void*, const void*, void*, void*, int,
+                             int, long, int, hipStream_t);
 }
 
 torch::Tensor lt_linear_gelu_bias(torch::Tensor, torch::Tensor,
@@ -417,6 +419,30 @@ std::vector<torch::Tensor> flash_bwd_fused(torch::Tensor q, torch::Tensor k,
   return {dk, dv};
 }
 
+torch::Tensor wgrad_gemm(torch::Tensor dy, torch::Tensor x,
+                         long splits) {
+  // dW[M, N] = dy[K, M]^T @ x[K, N]; custom split-K MFMA kernel
+  check_bf16(dy, "dy"); check_bf16(x, "x");
+  TORCH_CHECK(dy.dim() == 2 && x.dim() == 2, "2-D operands required");
+  const long K = dy.size(0), M = dy.size(1), N = x.size(1);
+  TORCH_CHECK(x.size(0) == K, "K mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0, "M, N must be 256-multiples");
+  TORCH_CHECK(K % 32 == 0, "K must be a 32-multiple");
+  int S = (int)splits;
+  if (S <= 0) {   // auto: enough workgroups to fill 256 CUs
+    long tiles = (M / 256) * (N / 256);
+    S = 1;
+    while (S < 32 && tiles * S < 512 && (K / (2L * S)) % 32 == 0) S *= 2;
+  }
+  TORCH_CHECK((K / S) % 32 == 0, "K/S must be a 32-multiple");
+  auto ws = torch::empty({S, M, N}, dy.options().dtype(torch::kFloat32));
+  auto out = torch::empty({M, N}, dy.options());
+  CHECK_HIP(wgrad_gemm_launch(dy.data_ptr(), x.data_ptr(), ws.data_ptr(),
+                              out.data_ptr(), (int)M, (int)N, K, S,
+                              cur_stream()));
+  return out;
+}
+
 torch::Tensor bias_grad(torch::Tensor dy) {
   check_bf16(dy, "dy");
   const int D = (int)dy.size(-1);
@@ -569,6 +595,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("emit_ds") = false);
   m.def("flash_dq_recompute", &flash_dq_recompute,
         "dQ by recompute: S/dP/dS in-register, no dS materialization");
+  m.def("wgrad_gemm", &wgrad_gemm,
+        "split-K wgrad GEMM: dW = dy^T @ x (custom MFMA)",
+        py::arg("dy"), py::arg("x"), py::arg("splits") = 0);
   m.def("flash_fwd_packed", &flash_fwd_packed,
         "flash fwd on the packed [B,L,3D] QKV projection output");
   m.def("flash_bwd_packed", &flash_bwd_packed,

@@ -31,6 +31,7 @@ ext = CUDAExtension(
         "csrc/adamw.hip",
         "csrc/repack.hip",
         "csrc/flash_attn.hip",
+        "csrc/wgrad_gemm.hip",
         "csrc/pool.hip",
     ],
     libraries=["hipblaslt"],
