@@ -28,6 +28,7 @@
 // SURVEY.md §2.1); this backs the MLTC classifier lane's training step.
 
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
@@ -217,6 +218,141 @@ wgrad_gemm_kernel(const short* __restrict__ a,   // [K, M] bf16
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2 (round 2): same tiling/compute, but staging via global_load_lds
+// (direct-to-LDS DMA) with DOUBLE-buffered tiles, raw s_barriers and
+// counted vmcnt waits — the guide's pipeline structure that the
+// register-staged 2-barrier loop cannot reach (its drain-everything
+// barrier caps the structure ~0.9 PF).  The XOR tile swizzle survives
+// glds' linear wave-order writes by PRE-SWIZZLING THE PER-LANE GLOBAL
+// ADDRESS (the swizzle is an XOR permutation of 16-B groups, so each
+// lane simply fetches the group that belongs at its linear LDS slot);
+// the tr-read compute path is unchanged and stays conflict-free.
+// vmcnt discipline: 4 glds per wave per tile (2 A + 2 B), one tile
+// prefetched ahead -> steady-state s_waitcnt vmcnt(4), drain 0 on last.
+
+__device__ __forceinline__ void wg_issue_glds(
+    const short* __restrict__ g, char* lds_tile, int wid, int lane,
+    long ktop, long ld, int base_col, int width) {
+  // wave w stages LDS bytes [w*2048, w*2048+2048) of the 16 KB tile as
+  // two 1024-B glds chunks; per chunk each lane contributes 16 B at
+  // lds + chunk_base + lane*16, fetching the swizzle-compensated group.
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int o = wid * 2048 + c * 1024 + lane * 16;
+    int block = o >> 12;
+    int ob = o & 4095;
+    int k = ob >> 7;
+    int s16 = (ob & 127) >> 4;
+    int m16 = s16 ^ (k & 7);
+    const short* gp = g + (ktop + k) * ld + base_col + (block * 8 + m16) * 8;
+    // issue via inline asm (M0 = wave-uniform LDS base; the instruction
+    // adds lane*16): hipcc's builtin path inserts a conservative
+    // s_waitcnt vmcnt(0) before every glds batch, draining the prefetch
+    // pipeline — the asm form keeps the compiler out of the loop.
+    unsigned lds_base = (unsigned)(unsigned long)
+        (lds_tile + wid * 2048 + c * 1024);
+    unsigned m0v = __builtin_amdgcn_readfirstlane(lds_base);
+    asm volatile(
+        "s_mov_b32 m0, %0\n\t"
+        "global_load_lds_dwordx4 %1, off"
+        :: "s"(m0v), "v"(gp));
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(WG_THREADS, 1)
+wgrad_gemm_glds_kernel(const short* __restrict__ a,   // [K, M] bf16
+                       const short* __restrict__ b,   // [K, N] bf16
+                       float* __restrict__ ws,        // [S, M, N] f32
+                       int M, int N, long K, int S) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // double-buffered: [2][A 16KB][B 16KB]
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int col = lane & 31;
+
+  const int tiles_n = (N + WG_BN - 1) / WG_BN;
+  int bid = (int)blockIdx.x;
+  const int tm = bid / tiles_n;
+  const int tn = bid % tiles_n;
+  const int split = blockIdx.y;
+  const long k_per = (K / S / WG_BK) * WG_BK;
+  const long k0 = (long)split * k_per;
+  const long k1 = (split == S - 1) ? K : k0 + k_per;
+  const int m_base = tm * WG_BM;
+  const int n_base = tn * WG_BN;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+  const int wm = m_base + wr * WG_WAVE_M;
+  const int wn = n_base + wc * WG_WAVE_N;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.f);
+
+  const long nt = (k1 - k0) / WG_BK;
+  wg_issue_glds(a, smem, wid, lane, k0, M, m_base, WG_BM);
+  wg_issue_glds(b, smem + WG_TILE_BYTES, wid, lane, k0, N, n_base, WG_BN);
+  for (long t = 0; t < nt; ++t) {
+    char* buf = smem + (t & 1) * (2 * WG_TILE_BYTES);
+    if (t + 1 < nt) {
+      char* nbuf = smem + ((t + 1) & 1) * (2 * WG_TILE_BYTES);
+      long ktop = k0 + (t + 1) * WG_BK;
+      wg_issue_glds(a, nbuf, wid, lane, ktop, M, m_base, WG_BM);
+      wg_issue_glds(b, nbuf + WG_TILE_BYTES, wid, lane, ktop, N, n_base,
+                    WG_BN);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    short8_t fa[2][2][2];
+    short8_t fb[2][2];
+    wg_tr_gather((unsigned)(unsigned long)(buf + (wr * 2 + 0) * WG_BLK_BYTES),
+                 lane, fa[0]);
+    wg_tr_gather((unsigned)(unsigned long)(buf + (wr * 2 + 1) * WG_BLK_BYTES),
+                 lane, fa[1]);
+    wg_tr_gather((unsigned)(unsigned long)(buf + WG_TILE_BYTES +
+                                           wc * WG_BLK_BYTES),
+                 lane, fb);
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        short8_t af = fa[mf >> 1][c][mf & 1];
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, fb[c][nf], acc[mf][nf], 0, 0, 0);
+        }
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const int half = lane >> 5;
+  float* wsp = ws + (long)split * M * N;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int ml = (r & 3) + 8 * (r >> 2) + 4 * half;
+        long row = wm + mf * 32 + ml;
+        long cc = wn + nf * 32 + col;
+        if (row < M && cc < N) wsp[row * N + cc] = acc[mf][nf][r];
+      }
+    }
+  }
+}
+
 // Deterministic split reduce: out[MN] bf16 = sum_s ws[s][MN]
 extern "C" __global__ void __launch_bounds__(256)
 wgrad_reduce_kernel(const float* __restrict__ ws, short* __restrict__ out,
@@ -241,9 +377,16 @@ extern "C" hipError_t wgrad_gemm_launch(const void* a, const void* b,
   int tiles_m = (M + WG_BM - 1) / WG_BM;
   int tiles_n = (N + WG_BN - 1) / WG_BN;
   dim3 grid(tiles_m * tiles_n, S);
-  size_t shm = 2 * WG_TILE_BYTES;
-  wgrad_gemm_kernel<<<grid, WG_THREADS, shm, stream>>>(
-      (const short*)a, (const short*)b, (float*)ws, M, N, K, S);
+  const char* v = getenv("TOSEM_WGRAD_V1");
+  if (v && v[0] == '1') {
+    size_t shm = 2 * WG_TILE_BYTES;
+    wgrad_gemm_kernel<<<grid, WG_THREADS, shm, stream>>>(
+        (const short*)a, (const short*)b, (float*)ws, M, N, K, S);
+  } else {
+    size_t shm = 4 * WG_TILE_BYTES;    // double-buffered glds variant
+    wgrad_gemm_glds_kernel<<<grid, WG_THREADS, shm, stream>>>(
+        (const short*)a, (const short*)b, (float*)ws, M, N, K, S);
+  }
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return e;
   long mn = (long)M * N;
