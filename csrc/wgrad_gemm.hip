@@ -46,6 +46,35 @@ __device__ __forceinline__ int wg_swz(int k, int byte_off) {
   return byte_off ^ ((k & 7) << 4);
 }
 
+// XCD-aware supertile decode (round 2): HBM traffic is the wall — each
+// 256-col A panel is re-read by tiles_n tile-columns and each B panel by
+// tiles_m rows (~3.2 GB vs the 0.54 GB ideal at the qkv shape).  Group
+// output tiles into 2(M) x 4(N) supertiles and give all 8 members ids
+// congruent mod 8 so the dispatcher (XCD = id % 8) co-locates them on one
+// XCD: the shared A/B panel k-windows then hit that XCD's L2 (A traffic
+// /4, B /2).  Grid x = SP_padded * 8 with SP padded to a multiple of 8
+// (members sit at x = sp + j*SP_padded); pad blocks exit immediately.
+// Requires tiles_m % 2 == 0 and tiles_n % 4 == 0 (host guarantees via the
+// 256-multiple shape check and falls back to plain ids otherwise).
+__device__ __forceinline__ bool wg_decode(int x, int tiles_m, int tiles_n,
+                                          int* tm, int* tn) {
+  if ((tiles_m & 1) || (tiles_n & 3)) {
+    if (x >= tiles_m * tiles_n) return false;
+    *tm = x / tiles_n;
+    *tn = x % tiles_n;
+    return true;
+  }
+  int real_sp = (tiles_m / 2) * (tiles_n / 4);
+  int sp_pad = (real_sp + 7) & ~7;
+  int sp = x % sp_pad;
+  int j = x / sp_pad;
+  if (sp >= real_sp) return false;
+  int stm = j >> 2, stn = j & 3;
+  *tm = (sp % (tiles_m / 2)) * 2 + stm;
+  *tn = (sp / (tiles_m / 2)) * 4 + stn;
+  return true;
+}
+
 // Gather the four [32 k] x [64 col] fragments of one LDS column-block via
 // ds_read_b64_tr_b16 (c = k-chunk 0/1, t = col-half 0/1), exactly the
 // flash kernels' V^T/K^T idiom.
@@ -111,10 +140,10 @@ wgrad_gemm_kernel(const short* __restrict__ a,   // [K, M] bf16
   const int wid = tid / WAVE;
   const int col = lane & 31;
 
+  const int tiles_m = (M + WG_BM - 1) / WG_BM;
   const int tiles_n = (N + WG_BN - 1) / WG_BN;
-  int bid = (int)blockIdx.x;
-  const int tm = bid / tiles_n;
-  const int tn = bid % tiles_n;
+  int tm, tn;
+  if (!wg_decode((int)blockIdx.x, tiles_m, tiles_n, &tm, &tn)) return;
   const int split = blockIdx.y;
   const long k_per = (K / S / WG_BK) * WG_BK;   // host guarantees exact
   const long k0 = (long)split * k_per;
@@ -272,10 +301,10 @@ wgrad_gemm_glds_kernel(const short* __restrict__ a,   // [K, M] bf16
   const int wid = tid / WAVE;
   const int col = lane & 31;
 
+  const int tiles_m = (M + WG_BM - 1) / WG_BM;
   const int tiles_n = (N + WG_BN - 1) / WG_BN;
-  int bid = (int)blockIdx.x;
-  const int tm = bid / tiles_n;
-  const int tn = bid % tiles_n;
+  int tm, tn;
+  if (!wg_decode((int)blockIdx.x, tiles_m, tiles_n, &tm, &tn)) return;
   const int split = blockIdx.y;
   const long k_per = (K / S / WG_BK) * WG_BK;
   const long k0 = (long)split * k_per;
@@ -376,7 +405,12 @@ extern "C" hipError_t wgrad_gemm_launch(const void* a, const void* b,
                                         long K, int S, hipStream_t stream) {
   int tiles_m = (M + WG_BM - 1) / WG_BM;
   int tiles_n = (N + WG_BN - 1) / WG_BN;
-  dim3 grid(tiles_m * tiles_n, S);
+  int gx = tiles_m * tiles_n;
+  if (tiles_m % 2 == 0 && tiles_n % 4 == 0) {
+    int real_sp = (tiles_m / 2) * (tiles_n / 4);
+    gx = ((real_sp + 7) & ~7) * 8;      // supertile-padded (wg_decode)
+  }
+  dim3 grid(gx, S);
   const char* v = getenv("TOSEM_WGRAD_V1");
   if (v && v[0] == '1') {
     size_t shm = 2 * WG_TILE_BYTES;
