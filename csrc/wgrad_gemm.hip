@@ -56,22 +56,17 @@ __device__ __forceinline__ int wg_swz(int k, int byte_off) {
 // (members sit at x = sp + j*SP_padded); pad blocks exit immediately.
 // Requires tiles_m % 2 == 0 and tiles_n % 4 == 0 (host guarantees via the
 // 256-multiple shape check and falls back to plain ids otherwise).
+// MEASURED NEGATIVE (round 2): a 2x4 supertile mapping with all 8 members
+// id-congruent mod 8 (same XCD, shared panel k-windows in its L2)
+// regressed EVERY shape — outp 174 -> 566 us, qkv 488 -> 590 — XCD
+// co-location serializes the members on one XCD's CUs and the padded
+// grid wastes dispatch slots; plain row-major tile order keeps panel
+// reuse in the chip-level L2 well enough.  Kept as the plain decode.
 __device__ __forceinline__ bool wg_decode(int x, int tiles_m, int tiles_n,
                                           int* tm, int* tn) {
-  if ((tiles_m & 1) || (tiles_n & 3)) {
-    if (x >= tiles_m * tiles_n) return false;
-    *tm = x / tiles_n;
-    *tn = x % tiles_n;
-    return true;
-  }
-  int real_sp = (tiles_m / 2) * (tiles_n / 4);
-  int sp_pad = (real_sp + 7) & ~7;
-  int sp = x % sp_pad;
-  int j = x / sp_pad;
-  if (sp >= real_sp) return false;
-  int stm = j >> 2, stn = j & 3;
-  *tm = (sp % (tiles_m / 2)) * 2 + stm;
-  *tn = (sp / (tiles_m / 2)) * 4 + stn;
+  if (x >= tiles_m * tiles_n) return false;
+  *tm = x / tiles_n;
+  *tn = x % tiles_n;
   return true;
 }
 
@@ -405,12 +400,7 @@ extern "C" hipError_t wgrad_gemm_launch(const void* a, const void* b,
                                         long K, int S, hipStream_t stream) {
   int tiles_m = (M + WG_BM - 1) / WG_BM;
   int tiles_n = (N + WG_BN - 1) / WG_BN;
-  int gx = tiles_m * tiles_n;
-  if (tiles_m % 2 == 0 && tiles_n % 4 == 0) {
-    int real_sp = (tiles_m / 2) * (tiles_n / 4);
-    gx = ((real_sp + 7) & ~7) * 8;      // supertile-padded (wg_decode)
-  }
-  dim3 grid(gx, S);
+  dim3 grid(tiles_m * tiles_n, S);
   const char* v = getenv("TOSEM_WGRAD_V1");
   if (v && v[0] == '1') {
     size_t shm = 2 * WG_TILE_BYTES;

@@ -93,12 +93,20 @@ class FusedLayerNorm(nn.Module):
 # as in round 1.  Default stays OFF; the ~1.4 ms of at::native::reduce
 # bias grads is the price of addmm's better GEMM dispatch.
 _USE_FUSED_LINEAR = os.environ.get("TOSEM_FUSED_LINEAR", "0") == "1"
+# TOSEM_WGRAD=1: weight grads of the big projections through the custom
+# split-K MFMA wgrad kernel (csrc/wgrad_gemm.hip) + colsum bias grad;
+# forward addmm and dgrad dispatch unchanged.
+_USE_WGRAD = os.environ.get("TOSEM_WGRAD", "0") == "1"
 
 
 def _linear(mod: nn.Linear, x: torch.Tensor) -> torch.Tensor:
-    if _USE_FUSED_LINEAR and x.is_cuda and mod.bias is not None \
-            and torch.is_grad_enabled():
-        return ops.fused_linear(x, mod.weight, mod.bias)
+    if x.is_cuda and torch.is_grad_enabled():
+        if _USE_WGRAD and ops.wgrad_linear_supported(
+                mod.out_features, mod.in_features,
+                x.numel() // x.shape[-1]):
+            return ops.wgrad_linear(x, mod.weight, mod.bias)
+        if _USE_FUSED_LINEAR and mod.bias is not None:
+            return ops.fused_linear(x, mod.weight, mod.bias)
     return mod(x)
 
 
@@ -152,7 +160,7 @@ class FFN(nn.Module):
             a = ops.lt_linear_gelu_bias(x.reshape(-1, D).contiguous(),
                                         self.up.weight, self.up_bias)
             return self.down(a.view(B, L, -1))
-        h = self.up(x)
+        h = _linear(self.up, x)
         h = ops.fused_bias_gelu(h, self.up_bias)
         return _linear(self.down, h)
 

@@ -250,6 +250,44 @@ def fused_linear(x, w, b):
     return y.view(*lead, -1)
 
 
+class _LinearWgradFn(torch.autograd.Function):
+    """Linear whose WEIGHT gradient runs through the custom split-K MFMA
+    wgrad kernel (csrc/wgrad_gemm.hip) and the bias gradient through the
+    colsum bias_grad kernel; forward is the same fused addmm GEMM, dgrad
+    the same torch matmul dispatch."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        if b is None:
+            return torch.matmul(x, w.t())
+        return torch.addmm(b, x, w.t())
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = torch.matmul(dy, w)
+        dw = hip_ops().wgrad_gemm(dy, x, 0)
+        db = hip_ops().bias_grad(dy) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def wgrad_linear_supported(out_features: int, in_features: int,
+                           tokens: int) -> bool:
+    return out_features % 256 == 0 and in_features % 256 == 0 and \
+        tokens % 32 == 0
+
+
+def wgrad_linear(x, w, b=None):
+    """Linear with the custom wgrad kernel (bf16 CUDA; 256-multiple
+    features, token count % 32)."""
+    lead = x.shape[:-1]
+    y = _LinearWgradFn.apply(x.reshape(-1, x.shape[-1]).contiguous(), w, b)
+    return y.view(*lead, -1)
+
+
 def lt_linear_gelu_bias(x, w1, b1):
     """GELU(x @ w1^T + b1) in one hipBLASLt GEMM (GELU_BIAS epilogue) —
     inference only: this hipBLASLt has no aux epilogues (no pre-activation
