@@ -662,17 +662,36 @@ colsum_bf16_stage2_kernel(const float* __restrict__ scratch,
   out[col] = f32_to_bf16(s);
 }
 
+// Adaptive splits (round 2): the fixed 64-way split left a [65536, 1024]
+// bias grad running on 64 blocks of a 256-CU chip — 452 us/call, which
+// was the REAL cause of the round-1 "fused_linear regresses 13.7 ms"
+// finding (misattributed to GEMM dispatch).  Stage 1 launches ~1024
+// blocks; a float4 middle stage collapses the splits to 16; the bf16
+// stage finishes.  scratch needs (1024 + 16) rows (ops.cpp).
 hipError_t colsum_bf16_launch(const void* x, void* scratch, void* out,
                               long N, int D, hipStream_t stream) {
-  const int splits = 64;
+  int gx = (D / 8 + 255) / 256;
+  int splits = 1024 / gx;
+  if (splits > 1024) splits = 1024;
+  if ((long)splits > N) splits = (int)N;
   long rows_per_split = (N + splits - 1) / splits;
-  dim3 g1((D / 8 + 255) / 256, splits);
+  dim3 g1(gx, splits);
   colsum_bf16_stage1_kernel<<<g1, 256, 0, stream>>>(
       (const short*)x, (float*)scratch, N, D, rows_per_split);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return e;
-  dim3 g2((D + 255) / 256);
-  colsum_bf16_stage2_kernel<<<g2, 256, 0, stream>>>(
+  if (splits > 16 && D % 4 == 0) {
+    float* mid = (float*)scratch + (long)1024 * D;
+    int rps2 = (splits + 15) / 16;
+    colsum_stage_kernel<<<dim3((D / 4 + 255) / 256, 16), 256, 0, stream>>>(
+        (const float*)scratch, mid, splits, D, rps2);
+    e = hipGetLastError();
+    if (e != hipSuccess) return e;
+    colsum_bf16_stage2_kernel<<<dim3((D + 255) / 256), 256, 0, stream>>>(
+        mid, (short*)out, 16, D);
+    return hipGetLastError();
+  }
+  colsum_bf16_stage2_kernel<<<dim3((D + 255) / 256), 256, 0, stream>>>(
       (const float*)scratch, (short*)out, splits, D);
   return hipGetLastError();
 }

@@ -448,7 +448,8 @@ torch::Tensor bias_grad(torch::Tensor dy) {
   const int D = (int)dy.size(-1);
   const long N = dy.numel() / D;
   TORCH_CHECK(D % 8 == 0, "D % 8 == 0 required");
-  auto scratch = torch::empty({64, D}, dy.options().dtype(torch::kFloat32));
+  auto scratch = torch::empty({1024 + 16, D},
+                              dy.options().dtype(torch::kFloat32));
   auto out = torch::empty({D}, dy.options());
   CHECK_HIP(colsum_bf16_launch(dy.data_ptr(), scratch.data_ptr(),
                                out.data_ptr(), N, D, cur_stream()));
