@@ -129,6 +129,16 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
     }
 
+    // mask-bias loads HOISTED above the QK chain (round 2): issued per
+    // register inside the softmax loop they sit exposed on the serial
+    // path between the mfma results and the exp chain
+    float mb_r[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+      mb_r[r] = mrow ? mrow[kv0 + kv_local] : 0.f;
+    }
+
     // ---- QK^T for BOTH q-blocks (8 back-to-back MFMAs) ----
     f32x16 sA = (f32x16)(0.f), sB = (f32x16)(0.f);
     __builtin_amdgcn_s_setprio(1);
@@ -147,8 +157,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     float tmaxA = -3.0e38f, tmaxB = -3.0e38f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float mb = mrow ? mrow[kv0 + kv_local] : 0.f;
+      float mb = mb_r[r];
       float xA = sA[r] * scale + mb;
       float xB = sB[r] * scale + mb;
       svA[r] = xA; svB[r] = xB;
@@ -872,6 +881,13 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
                                (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
     }
 
+    float mb_r[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
+      mb_r[r] = mrow ? mrow[kv0 + kv_local] : 0.f;
+    }
+
     // ---- S and dP for BOTH q-blocks (16 back-to-back MFMAs) ----
     f32x16 sA = (f32x16)(0.f), sB = (f32x16)(0.f);
     f32x16 pA = (f32x16)(0.f), pB = (f32x16)(0.f);
@@ -894,10 +910,8 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
     float gA[16], gB[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      int kv_local = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float mb = mrow ? mrow[kv0 + kv_local] : 0.f;
-      float expA = __expf(sA[r] * scale + mb - lseA);
-      float expB = __expf(sB[r] * scale + mb - lseB);
+      float expA = __expf(sA[r] * scale + mb_r[r] - lseA);
+      float expB = __expf(sB[r] * scale + mb_r[r] - lseB);
       gA[r] = scale * expA * (pA[r] - ddA);
       gB[r] = scale * expB * (pB[r] - ddB);
     }
