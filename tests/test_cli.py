@@ -76,3 +76,43 @@ def test_train_then_classify_cli(mini_corpus, tmp_path):
     from tosem2021_amd.analyze.taxonomy import load_taxonomy
     df = load_taxonomy(out)
     assert len(df) > 0
+
+
+def test_cli_labels_command(tmp_path, capsys):
+    """labels subcommand: L2 ingestion + lineage report."""
+    import json
+    import os
+
+    import pytest
+
+    if not os.path.isdir("/root/reference/selection/completed-labels"):
+        pytest.skip("reference corpus not mounted on this box")
+    from tosem2021_amd.cli import main
+    out = str(tmp_path / "labels.json")
+    main(["labels", "--taxonomy", "/root/reference/RQs/taxonomy_test2.csv",
+          "--json", out])
+    with open(out) as f:
+        res = json.load(f)
+    assert res["codebook"]["strategies_uncovered"] == []
+    assert len(res["release_sheets"]) == 8
+    assert all(v["coverage"] == 1.0 for v in res["lineage"].values())
+
+
+def test_cli_golden_mirror(tmp_path):
+    """analyze (mirror) + golden --mirror round trip on the reference."""
+    import os
+
+    import pytest
+
+    if not os.path.exists("/root/reference/RQs/taxonomy_test2.csv"):
+        pytest.skip("reference corpus not mounted on this box")
+    from tosem2021_amd.cli import main
+    out = str(tmp_path / "RQs")
+    main(["analyze", "--taxonomy", "/root/reference/RQs/taxonomy_test2.csv",
+          "--out", out])
+    assert os.path.exists(os.path.join(out, "RQ3",
+                                       "tests_correlate_assertion.csv"))
+    with pytest.raises(SystemExit) as exc:
+        main(["golden", "--ours", out, "--reference",
+              "/root/reference/RQs", "--mirror"])
+    assert exc.value.code == 0
