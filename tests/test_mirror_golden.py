@@ -111,3 +111,32 @@ def test_mirror_runs_on_mined_style_taxonomy(tmp_path):
     assert len(paths) >= 26
     for p in paths.values():
         assert os.path.getsize(p) > 0
+
+
+def test_mirror_self_consistency(tmp_path):
+    """golden_mirror comparing a mirror against ITSELF: every file ok with
+    Pearson 1.0 — guards the comparator and the emitters jointly."""
+    import pandas as pd
+
+    from tosem2021_amd.analyze.golden_mirror import mirror_diff
+    from tosem2021_amd.analyze.mirror import write_mirror
+    from tosem2021_amd.extract.schema import TAXONOMY_COLUMNS
+    rows = []
+    for i in range(60):
+        row = {c: "" for c in TAXONOMY_COLUMNS}
+        row.update({"Index": i, "Labels": f"assertEqual(x, {i})",
+                    "Category": "model_training", "Cases": 1, "FileID": i,
+                    "Repo": ["Apollo", "Ray", "nni"][i % 3],
+                    "Data": "Data Validity" if i % 2 else "",
+                    "status_test": i % 2, "negative_test": (i + 1) % 2,
+                    "error_handling": i % 3 == 0,
+                    "Error_Type": "ValueError" if i % 3 == 0 else ""})
+        rows.append(row)
+    df = pd.DataFrame(rows, columns=TAXONOMY_COLUMNS)
+    out = str(tmp_path / "m")
+    write_mirror(df, out)
+    res = mirror_diff(out, out)
+    assert res["ok"], res
+    for rel, v in res["files"].items():
+        if "pearson" in v and v.get("cells", 0) > 2:
+            assert v["pearson"] > 0.999, (rel, v)
