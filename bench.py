@@ -39,6 +39,19 @@ elif os.path.exists(_TUNE_FILE.replace(".csv", "0.csv")) and \
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNE_FILE)
+    # TunableOp resolves the table per DEVICE ordinal (…gfx950<K>.csv); on
+    # a multi-GPU launch ranks 1..7 would otherwise find no table and run
+    # the untuned heuristics (dragging the max-over-ranks step time) —
+    # replicate the tuned table for every local ordinal.
+    try:
+        _src = _TUNE_FILE.replace(".csv", "0.csv")
+        for _k in range(1, 8):
+            _dst = _TUNE_FILE.replace(".csv", f"{_k}.csv")
+            if not os.path.exists(_dst):
+                import shutil
+                shutil.copyfile(_src, _dst)
+    except OSError:
+        pass   # read-only checkout: ordinal-0 rank still gets the table
 
 import torch
 
