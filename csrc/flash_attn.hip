@@ -181,16 +181,23 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     lA = lA * aA + rsA; mA = mnA;
     lB = lB * aB + rsB; mB = mnB;
 
-    // alpha broadcast for both blocks (two slots per wave)
-    alpha_lds[wid * 64 + col] = aA;
-    alpha_lds[wid * 64 + 32 + col] = aB;
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
+    // O rescale (round 2): alpha broadcast by __shfl (v_readlane — the
+    // value for q-row r lives in lane r) instead of an LDS round-trip
+    // (was 2 stores + 64 reads/tile), and skipped wave-uniformly when no
+    // row's running max moved this tile (aA==aB==1 exactly; common once
+    // the max stabilizes a few tiles in).  Wave-local state only, so the
+    // uniform skip is race-free.
+    if (__builtin_amdgcn_ballot_w64(aA != 1.f || aB != 1.f)) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        oA[t][r] *= alpha_lds[wid * 64 + qrow];
-        oB[t][r] *= alpha_lds[wid * 64 + 32 + qrow];
+        float alA = __shfl(aA, qrow, WAVE);
+        float alB = __shfl(aB, qrow, WAVE);
+#pragma unroll
+        for (int t = 0; t < 2; ++t) {
+          oA[t][r] *= alA;
+          oB[t][r] *= alB;
+        }
       }
     }
 
@@ -284,20 +291,21 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
     if (validA) lse[(long)bh * L + my_qA] = mA + __logf(lA);
     if (validB) lse[(long)bh * L + my_qB] = mB + __logf(lB);
   }
-  alpha_lds[wid * 64 + col] = 1.0f / lA;
-  alpha_lds[wid * 64 + 32 + col] = 1.0f / lB;
+  const float invA = 1.0f / lA, invB = 1.0f / lB;
 #pragma unroll
-  for (int t = 0; t < 2; ++t) {
+  for (int r = 0; r < 16; ++r) {
+    int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
+    float iA = __shfl(invA, rloc, WAVE);
+    float iB = __shfl(invB, rloc, WAVE);
+    int qA = q_baseA + rloc, qB = q_baseB + rloc;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      int rloc = (r & 3) + 8 * (r >> 2) + 4 * half;
-      int qA = q_baseA + rloc, qB = q_baseB + rloc;
+    for (int t = 0; t < 2; ++t) {
       if (qA < L)
         o[o_off + (long)qA * o_rs + 32 * t + col] =
-            f32_to_bf16(oA[t][r] * alpha_lds[wid * 64 + rloc]);
+            f32_to_bf16(oA[t][r] * iA);
       if (qB < L)
         o[o_off + (long)qB * o_rs + 32 * t + col] =
-            f32_to_bf16(oB[t][r] * alpha_lds[wid * 64 + 32 + rloc]);
+            f32_to_bf16(oB[t][r] * iB);
     }
   }
 }
