@@ -52,6 +52,27 @@ __device__ __forceinline__ int kswz(int row, int byte_off) {
 // attention output written straight into the [B, L, D] proj input
 // (o_bs=L*D, o_hs=64, o_rs=D).  The packed path deletes the four
 // qkv/out repack kernels (+their backward merges) from the model step.
+
+// Padding-tail tile skip (round 2): with the additive -1e9 padding bias,
+// every score in a fully-masked kv tile sits below ~-1e8 and __expf
+// underflows to exactly +0.0f, so the tile contributes NOTHING to the
+// online softmax (alpha==1, rowsum+=0, O+=0) or to dQ — identical to not
+// running it.  The mask is per-(batch, kv) so the last unmasked position
+// is workgroup-uniform; tiles past it are skipped.  If the whole row is
+// masked (last==-1) nothing is skipped, preserving the reference
+// softmax-over-bias behavior for degenerate all-pad sequences.
+__device__ __forceinline__ int fa_last_active_kv(const float* mrow, int L,
+                                                 int tid, int* scratch) {
+  if (tid == 0) *scratch = -1;
+  __syncthreads();
+  int loc = -1;
+  for (int kv = tid; kv < L; kv += FA_BLOCK)
+    if (mrow[kv] > -1.0e8f) loc = kv;
+  if (loc >= 0) atomicMax(scratch, loc);
+  __syncthreads();
+  return *scratch;
+}
+
 extern "C" __global__ void __launch_bounds__(FA_BLOCK, 2)
 flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
                  const short* __restrict__ v, const float* __restrict__ mask,
@@ -109,12 +130,17 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
   float mA = -3.0e38f, lA = 0.f, mB = -3.0e38f, lB = 0.f;
 
   const int n_kv = L / FA_KVB;
+  int n_kv_eff = n_kv;
+  if (mrow) {
+    int last = fa_last_active_kv(mrow, L, tid, (int*)alpha_lds);
+    if (last >= 0) n_kv_eff = min(n_kv, last / FA_KVB + 1);
+  }
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
   short8_t kv8 = *(const short8_t*)(k + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
   short8_t vv8 = *(const short8_t*)(v + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
-  for (int kt = 0; kt < n_kv; ++kt) {
+  for (int kt = 0; kt < n_kv_eff; ++kt) {
     const int kv0 = kt * FA_KVB;
     __syncthreads();
     {
@@ -122,7 +148,7 @@ flash_fwd_kernel(const short* __restrict__ q, const short* __restrict__ k,
       *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
     }
     __syncthreads();
-    if (kt + 1 < n_kv) {
+    if (kt + 1 < n_kv_eff) {
       kv8 = *(const short8_t*)(k + qkv_off +
                                (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
       vv8 = *(const short8_t*)(v + qkv_off +
@@ -414,9 +440,29 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const float* mrow = mask ? mask + (long)b * L : nullptr;
   const float mb = mrow ? mrow[min(my_kv, L - 1)] : 0.f;
 
+  // Early-out (round 2): if every kv row this workgroup owns is padding
+  // (bias <= -1e8) and the sequence has at least one real token, P and dS
+  // are exactly +0.0f for every q — dK/dV are zero and the whole q loop is
+  // skipped (the epilogue writes the zero accumulators).  Requires
+  // ds == nullptr (the default path) so no dS tile is left unwritten; the
+  // mrow[0] guard keeps degenerate all-pad sequences on the reference
+  // softmax-over-bias behavior.  skip is workgroup-uniform (the flag is a
+  // barrier-ordered workgroup reduction), so branching around the
+  // barriered q loop is safe.
+  bool skip_tile = false;
+  if (mrow && ds == nullptr) {
+    int* any_lds = (int*)lse_lds;
+    if (tid == 0) *any_lds = 0;
+    __syncthreads();
+    if (kv_valid && mb > -1.0e8f) *any_lds = 1;
+    __syncthreads();
+    skip_tile = (*any_lds == 0) && (mrow[0] > -1.0e8f);
+    __syncthreads();
+  }
+
   // K and V fragments for this wave's kv block (resident all kernel)
   short8_t kf[4], vf[4];
-  {
+  if (!skip_tile) {
     const short* kr = k + qkv_off + (long)(kv_valid ? my_kv : L - 1) * qkv_rs;
     const short* vr = v + qkv_off + (long)(kv_valid ? my_kv : L - 1) * qkv_rs;
 #pragma unroll
@@ -431,7 +477,7 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
     dv_acc[t] = (f32x16)(0.f);
     dk_acc[t] = (f32x16)(0.f);
   }
-
+  if (!skip_tile) {
   const int n_q = L / 32;
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
   short8_t qv8 = *(const short8_t*)(q + qkv_off + (long)srow * qkv_rs +
@@ -603,6 +649,7 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
       __builtin_amdgcn_s_setprio(0);
     }
   }
+  }   // !skip_tile
 
   // epilogue: dV/dK rows of this wave's kv block (reg=kv row, lane=d col)
 #pragma unroll
@@ -872,12 +919,17 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
   for (int t = 0; t < 2; ++t) { dqA[t] = (f32x16)(0.f); dqB[t] = (f32x16)(0.f); }
 
   const int n_kv = L / FA_KVB;
+  int n_kv_eff = n_kv;
+  if (mrow) {
+    int last = fa_last_active_kv(mrow, L, tid, (int*)(smem + K_LDS_BYTES + VT_LDS_BYTES));
+    if (last >= 0) n_kv_eff = min(n_kv, last / FA_KVB + 1);
+  }
   const int srow = tid >> 3, sc8 = (tid & 7) * 16;
   short8_t kv8 = *(const short8_t*)(k + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
   short8_t vv8 = *(const short8_t*)(v + qkv_off + (long)srow * qkv_rs +
                                     (sc8 >> 1));
-  for (int kt = 0; kt < n_kv; ++kt) {
+  for (int kt = 0; kt < n_kv_eff; ++kt) {
     const int kv0 = kt * FA_KVB;
     __syncthreads();
     {
@@ -885,7 +937,7 @@ flash_dq_recompute_kernel(const short* __restrict__ q,
       *(short8_t*)((char*)v_lds + srow * 128 + kswz(srow, sc8)) = vv8;
     }
     __syncthreads();
-    if (kt + 1 < n_kv) {
+    if (kt + 1 < n_kv_eff) {
       kv8 = *(const short8_t*)(k + qkv_off +
                                (long)(kv0 + FA_KVB + srow) * qkv_rs + (sc8 >> 1));
       vv8 = *(const short8_t*)(v + qkv_off +
@@ -1035,7 +1087,7 @@ extern "C" hipError_t flash_dq_recompute_launch(
     int B, int H, int L, float scale, hipStream_t stream) {
   int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_qblocks);
-  size_t shm = K_LDS_BYTES + VT_LDS_BYTES;
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + 16;  // + tail-skip scratch
   flash_dq_recompute_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (const short*)dout,
       (const float*)mask, (const float*)lse, (const float*)ddot, (short*)dq,
@@ -1052,7 +1104,7 @@ extern "C" hipError_t flash_dq_recompute_packed_launch(
   const int D = H * FA_DH;
   int n_qblocks = (L + 2 * FA_QWG - 1) / (2 * FA_QWG);
   dim3 grid(B * H * n_qblocks);
-  size_t shm = K_LDS_BYTES + VT_LDS_BYTES;
+  size_t shm = K_LDS_BYTES + VT_LDS_BYTES + 16;  // + tail-skip scratch
   flash_dq_recompute_kernel<<<grid, FA_BLOCK, shm, stream>>>(
       (const short*)qkv, (const short*)qkv + D, (const short*)qkv + 2 * D,
       (const short*)dout, (const float*)mask, (const float*)lse,
