@@ -450,6 +450,7 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // barrier-ordered workgroup reduction), so branching around the
   // barriered q loop is safe.
   bool skip_tile = false;
+  bool wave_skip = false;
   if (mrow && ds == nullptr) {
     int* any_lds = (int*)lse_lds;
     if (tid == 0) *any_lds = 0;
@@ -458,6 +459,13 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
     __syncthreads();
     skip_tile = (*any_lds == 0) && (mrow[0] > -1.0e8f);
     __syncthreads();
+    // Finer grain: a wave whose OWN 32 kv rows are all padding still has
+    // to stage q/dO tiles and hit the barriers with the other waves, but
+    // its S/dP MFMA chains, softmax/dS math and dK/dV accumulation are
+    // exact zeros — skip them (wave-uniform ballot, no barriers inside
+    // the skipped region).
+    wave_skip = !skip_tile && (mrow[0] > -1.0e8f) &&
+        __builtin_amdgcn_ballot_w64(kv_valid && mb > -1.0e8f) == 0;
   }
 
   // K and V fragments for this wave's kv block (resident all kernel)
@@ -503,6 +511,7 @@ flash_bwd_fused_kernel(const short* __restrict__ q, const short* __restrict__ k,
                                (sc8 >> 1));
     }
 
+    if (wave_skip) continue;
     // S[q, kv] and dP[q, kv]: reg=q rows, lane=kv cols
     f32x16 s_acc = (f32x16)(0.f);
     f32x16 dp_acc = (f32x16)(0.f);
