@@ -129,10 +129,15 @@ def _load_dataset(taxonomy_path: str, cfg: MLTCConfig) -> TaxonomyDataset:
 def _run_steps(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
                batch: int, seq: int, steps: int, seed: int, device,
                eval_every: int = 0, val_ds: Optional[TaxonomyDataset] = None,
-               tag: str = "") -> list:
+               tag: str = "", augment: float = 0.0) -> list:
     losses = []
     epoch = 0
     it = iter(())
+    aug_keep = aug_gen = None
+    if augment > 0:
+        from tosem2021_amd.data.augment import augment_tokens, protected_ids
+        aug_keep = protected_ids(tok)
+        aug_gen = torch.Generator().manual_seed(seed * 7919 + 13)
     while trainer.step_num < steps:
         try:
             toks, mask, labels = next(it)
@@ -141,6 +146,10 @@ def _run_steps(trainer: Trainer, ds: TaxonomyDataset, tok: CodeTokenizer,
                             seed=seed + epoch, drop_last=True)
             epoch += 1
             continue
+        if augment > 0:
+            from tosem2021_amd.data.augment import augment_tokens
+            toks = augment_tokens(toks, augment, aug_keep,
+                                  tok.vocab_size, aug_gen)
         losses.append(trainer.step(toks, mask, labels))
         if eval_every and val_ds is not None and \
                 trainer.step_num % eval_every == 0:
@@ -162,7 +171,9 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
                      pretrain_steps: int = 0,
                      focal_gamma_property: float = 0.0,
                      label_smoothing: float = 0.0,
-                     dump_probs_path: Optional[str] = None) -> dict:
+                     dump_probs_path: Optional[str] = None,
+                     augment: float = 0.0,
+                     split_seed: Optional[int] = None) -> dict:
     """Fine-tune MLTC on `taxonomy_path`'s labeled rows.
 
     With `pretrain_path`/`pretrain_steps`, first train on that (typically
@@ -178,7 +189,11 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
     cfg = MLTCConfig(**{**base.__dict__, "max_seq": seq,
                      "dropout": dropout})
     full = _load_dataset(taxonomy_path, cfg)
-    train_ds, val_ds = full.split(val_frac=0.1, seed=seed)
+    # split_seed pins the train/val partition independently of the training
+    # seed, so differently-seeded runs share one val set and their dumped
+    # probs can be ensembled (docs/ROADMAP.md: shared-split k-fold)
+    train_ds, val_ds = full.split(
+        val_frac=0.1, seed=seed if split_seed is None else split_seed)
     tok = CodeTokenizer(cfg.vocab_size)
     tcfg = TrainConfig(model=model, lr=lr, warmup_steps=min(50, steps // 10),
                        total_steps=steps, ckpt_dir=ckpt_dir,
@@ -213,7 +228,8 @@ def train_classifier(taxonomy_path: str, model: str = "mltc-base",
         pretrain_time = time.time() - t0
 
     losses, epoch = _run_steps(trainer, train_ds, tok, batch, seq, steps,
-                               seed, dev, eval_every, val_ds)
+                               seed, dev, eval_every, val_ds,
+                               augment=augment)
     train_time = time.time() - t0
     ev = evaluate(trainer, val_ds, tok, seq)
     ev_lo = evaluate(trainer, val_ds, tok, seq, threshold=0.3)

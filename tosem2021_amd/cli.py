@@ -141,7 +141,8 @@ def cmd_train(args):
         pretrain_path=args.pretrain, pretrain_steps=args.pretrain_steps,
         focal_gamma_property=args.focal_gamma_property,
         label_smoothing=args.label_smoothing, eval_every=args.eval_every,
-        seed=args.seed, dump_probs_path=args.dump_probs)
+        seed=args.seed, dump_probs_path=args.dump_probs,
+        augment=args.augment, split_seed=args.split_seed)
     print(json.dumps(res, indent=2))
 
 
@@ -226,6 +227,12 @@ def main(argv=None):
     p.add_argument("--pretrain", default=None,
                    help="mined taxonomy to pretrain on before fine-tuning")
     p.add_argument("--pretrain-steps", type=int, default=0)
+    p.add_argument("--augment", type=float, default=0.0,
+                   help="fraction of non-protected token ids consistently "
+                        "renamed per example (assertion-text augmentation)")
+    p.add_argument("--split-seed", type=int, default=None,
+                   help="pin the train/val split independently of --seed "
+                        "(shared-split ensembling)")
     p.add_argument("--focal-gamma-property", type=float, default=0.0)
     p.add_argument("--label-smoothing", type=float, default=0.0)
     p.add_argument("--eval-every", type=int, default=0)
