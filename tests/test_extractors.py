@@ -215,8 +215,9 @@ it('next', () => { expect(y).to.equal(2); });
 
 
 def test_gtest_instantiate_multiplicity(tmp_path):
-    """TEST_P cases carry the INSTANTIATE_* count as param multiplicity
-    (each instantiation runs the whole suite once)."""
+    """TEST_P cases carry the summed static cardinality of their
+    INSTANTIATE_* value lists as param multiplicity (Values(1,2) +
+    Values(3) = 3 runs of each case)."""
     from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
     src = """
 class ParamSuite : public ::testing::TestWithParam<int> {};
@@ -237,11 +238,32 @@ INSTANTIATE_TEST_CASE_P(Legacy, ParamSuite, ::testing::Values(3));
     cases = {c.name: c for c in extract_gtest_file(str(p))}
     assert set(cases) == {"Works", "AlsoWorks", "One"}
     assert cases["Works"].is_parametrized
-    assert cases["Works"].param_multiplicity == 2
-    assert cases["AlsoWorks"].param_multiplicity == 2
+    assert cases["Works"].param_multiplicity == 3
+    assert cases["AlsoWorks"].param_multiplicity == 3
     assert not cases["One"].is_parametrized
     assert cases["One"].param_multiplicity == 1
     # classify_case folds the multiplicity into the Cases column
     from tosem2021_amd.classify.rules import classify_case
     rows = classify_case(cases["Works"], repo="Apollo", file_id=1)
-    assert rows[0].cases == 2     # 1 assertion x 2 instantiations
+    assert rows[0].cases == 3     # 1 assertion x 3 parameter values
+
+
+def test_gtest_instantiate_value_cardinality(tmp_path):
+    """INSTANTIATE_* generators are statically counted: Values argc,
+    Bool 2, Range span, Combine product; unknown generators fall back to
+    1 per instantiation."""
+    src = """
+TEST_P(MySuite, Works) { EXPECT_EQ(GetParam(), GetParam()); }
+INSTANTIATE_TEST_SUITE_P(Three, MySuite, testing::Values(1, 2, 3));
+INSTANTIATE_TEST_SUITE_P(Four, MySuite,
+    testing::Combine(testing::Values("a", "b"), testing::Bool()));
+
+TEST_P(Opaque, Runs) { EXPECT_TRUE(GetParam()); }
+INSTANTIATE_TEST_SUITE_P(Dyn, Opaque, testing::ValuesIn(kRuntimeVec));
+"""
+    p = tmp_path / "card_test.cc"
+    p.write_text(src)
+    from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+    cases = {c.name: c for c in extract_gtest_file(str(p))}
+    assert cases["Works"].param_multiplicity == 3 + 4
+    assert cases["Runs"].param_multiplicity == 1

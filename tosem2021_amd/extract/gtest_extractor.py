@@ -95,6 +95,81 @@ def _extract_call(text: str, start: int) -> str:
     return text[start:min(start + 200, n)]
 
 
+def _split_args(s: str) -> List[str]:
+    """Split a C++ argument list at top-level commas (tracks (), {}, [],
+    strings, and template <> with <</>>/comparison guards)."""
+    out, depth, ang, i, last, n = [], 0, 0, 0, 0, len(s)
+    while i < n:
+        c = s[i]
+        if c == '"':
+            i += 1
+            while i < n and s[i] != '"':
+                i += 2 if s[i] == "\\" else 1
+        elif c in "([{":
+            depth += 1
+        elif c in ")]}":
+            depth -= 1
+        elif c == "<":
+            if i + 1 < n and s[i + 1] in "<=":
+                i += 1
+            else:
+                ang += 1
+        elif c == ">":
+            if i + 1 < n and s[i + 1] in ">=":
+                i += 1
+            elif ang > 0 and not (i > 0 and s[i - 1] == "-"):
+                ang -= 1
+        elif c == "," and depth == 0 and ang == 0:
+            out.append(s[last:i].strip())
+            last = i + 1
+        i += 1
+    tail = s[last:].strip()
+    if tail or out:
+        out.append(tail)
+    return out
+
+
+_RE_GEN_HEAD = re.compile(
+    r"^(?:::)?(?:testing::|::testing::)?(Values|ValuesIn|Bool|Range|"
+    r"Combine|ConvertGenerator)\s*\(")
+
+
+def _gen_cardinality(expr: str) -> Optional[int]:
+    """Statically count the test cases an INSTANTIATE_* generator expands
+    to: Values(...) = argc, Bool() = 2, Range(b, e[, s]) from int literals,
+    Combine(...) = product.  None when unknowable (ValuesIn over a runtime
+    container, function calls, ...)."""
+    expr = expr.strip()
+    m = _RE_GEN_HEAD.match(expr)
+    if not m:
+        return None
+    head = m.group(1)
+    inner = _extract_call(expr, m.end() - 1)
+    args = _split_args(inner[1:-1]) if len(inner) >= 2 else []
+    if head == "Bool":
+        return 2
+    if head == "Values":
+        return len(args) or None
+    if head == "Range":
+        try:
+            b, e = int(args[0], 0), int(args[1], 0)
+            step = int(args[2], 0) if len(args) > 2 else 1
+            return max((e - b + step - 1) // step, 0) or None
+        except (ValueError, IndexError):
+            return None
+    if head == "Combine":
+        total = 1
+        for a in args:
+            c = _gen_cardinality(a)
+            if c is None:
+                return None
+            total *= c
+        return total
+    if head == "ConvertGenerator":
+        return _gen_cardinality(args[0]) if args else None
+    return None
+
+
 def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     rel = rel or path
     try:
@@ -131,7 +206,17 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     # missing item 6): each INSTANTIATE_* runs the whole suite once more
     inst_counts: dict = {}
     for im in RE_INSTANTIATE.finditer(text):
-        inst_counts[im.group(2)] = inst_counts.get(im.group(2), 0) + 1
+        # static value-list cardinality (docs/ROADMAP.md completeness
+        # item): Values/Bool/Range/Combine counted, else 1 per
+        # instantiation (conservative)
+        card = None
+        lp = text.find("(", im.start())
+        if lp >= 0:
+            call_args = _split_args(_extract_call(text, lp)[1:-1])
+            if len(call_args) >= 3:
+                card = _gen_cardinality(call_args[2])
+        inst_counts[im.group(2)] = \
+            inst_counts.get(im.group(2), 0) + (card or 1)
     marks = []
     for m in RE_TEST_MACRO.finditer(text):
         marks.append((m, m.group(2), m.group(3)))
