@@ -267,3 +267,47 @@ INSTANTIATE_TEST_SUITE_P(Dyn, Opaque, testing::ValuesIn(kRuntimeVec));
     cases = {c.name: c for c in extract_gtest_file(str(p))}
     assert cases["Works"].param_multiplicity == 3 + 4
     assert cases["Runs"].param_multiplicity == 1
+
+
+def test_freestanding_check_fallback(tmp_path):
+    """C++ test files with no framework macro (openfst style) fall back to
+    per-function cases over glog CHECK-family assertions."""
+    src = """
+#include <fst/float-weight.h>
+namespace {
+void TestWeightCopy(int w) {
+  CHECK_EQ(w, w);
+  CHECK(w >= 0);
+}
+template <class W>
+void TestNear(W a, W b) {
+  CHECK_NEAR(a, b, 1e-5);
+}
+void Helper(int x) { use(x); }   // no assertions -> no case
+}
+int main() { TestWeightCopy(3); return 0; }
+"""
+    p = tmp_path / "weight_test.cc"
+    p.write_text(src)
+    from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+    cases = {c.name: c for c in extract_gtest_file(str(p))}
+    assert "TestWeightCopy" in cases and "TestNear" in cases
+    assert "Helper" not in cases
+    assert len(cases["TestWeightCopy"].assertions) == 2
+    assert cases["TestNear"].assertions[0].kind == "approx"
+
+
+def test_walker_admits_tester_headers():
+    """Tester headers inside test dirs are walked (the study labeled
+    openfst's algo_test.h / weight-tester.h as DeepSpeech components);
+    ordinary headers are not."""
+    from tosem2021_amd.corpus.walker import classify_language, is_test_file
+    assert classify_language("a/b/weight-tester.h") == "cpp"
+    assert is_test_file("src/include/fst/test/weight-tester.h", "cpp")
+    assert is_test_file("src/include/fst/test/algo_test.h", "cpp")
+    assert not is_test_file("src/include/fst/float-weight.h", "cpp")
+    assert not is_test_file("src/lib/weight-tester.h", "cpp")   # not in test dir
+    # third_party is mined (study scope); node_modules still skipped
+    from tosem2021_amd.corpus.walker import SKIP_DIRS
+    assert "third_party" not in SKIP_DIRS
+    assert "node_modules" in SKIP_DIRS

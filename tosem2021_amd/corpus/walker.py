@@ -12,9 +12,19 @@ from typing import Iterator, List, Optional, Sequence
 
 PY_EXT = (".py",)
 CPP_EXT = (".cc", ".cpp", ".cxx")
+# headers are walked too, but is_test_file admits only tester headers in
+# test dirs (openfst's include/fst/test/{algo_test,fst_test,weight-tester}.h
+# back the study's biggest DeepSpeech components)
+CPP_HDR_EXT = (".h", ".hh", ".hpp")
 TS_EXT = (".ts", ".tsx")
 
-SKIP_DIRS = {".git", "node_modules", "third_party", "__pycache__", "build",
+# NOTE: third_party is NOT skipped — the study's taxonomy labels
+# DeepSpeech's vendored openfst suite (components "FST Algo Test",
+# "WeightTester", ... ~450 rows under native_client/ctcdecode/
+# third_party/openfst-*/src/test/); DeepSpeech is the only corpus
+# project with test files under a third_party dir, so including it is
+# scope-neutral for the other eight.
+SKIP_DIRS = {".git", "node_modules", "__pycache__", "build",
              "dist", ".tox", "external"}
 
 
@@ -29,7 +39,7 @@ class SourceFile:
 def classify_language(path: str) -> Optional[str]:
     if path.endswith(PY_EXT):
         return "python"
-    if path.endswith(CPP_EXT):
+    if path.endswith(CPP_EXT) or path.endswith(CPP_HDR_EXT):
         return "cpp"
     if path.endswith(TS_EXT):
         return "ts"
@@ -48,8 +58,13 @@ def is_test_file(rel: str, language: str) -> bool:
                             and not base.startswith("__"))
     if language == "cpp":
         stem = base.rsplit(".", 1)[0]
+        if base.endswith(CPP_HDR_EXT):
+            # tester headers only, and only inside a test dir
+            return in_test_dir and (stem.endswith("_test") or
+                                    stem.endswith("-tester") or
+                                    stem.endswith("_tester"))
         return stem.endswith("_test") or stem.endswith("_unittest") or \
-            stem.startswith("test_") or (in_test_dir and not base.endswith(".h"))
+            stem.startswith("test_") or in_test_dir
     if language == "ts":
         return ".test." in base or ".spec." in base
     return False

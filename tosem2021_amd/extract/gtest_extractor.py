@@ -33,6 +33,19 @@ RE_DEATH = re.compile(r"DEATH|THROW", re.I)
 # (reference src/DeepSpeech/v0.9.3/native_client/kenlm/lm/model_test.cc:11):
 # one-level expansion — a #define in the SAME file whose body contains a
 # known assertion macro makes its invocations count as that assertion.
+# glog/openfst-style CHECK assertions for FREESTANDING test files (no
+# TEST/BOOST macro; e.g. the vendored openfst suite the study labeled as
+# DeepSpeech components "FST Algo Test" / "WeightTester" — reference
+# src/DeepSpeech/v0.9.3/native_client/ctcdecode/third_party/openfst-*/
+# src/test/*.cc).  Only used on files where no framework macro matched,
+# so gtest files keep their EXPECT/ASSERT-only accounting.
+RE_CHECK = re.compile(r"\b((?:D|Q)?CHECK(?:_[A-Z_0-9]+)?)\s*\(")
+# a top-level C++ function definition heading a freestanding test body
+RE_FUNC_DEF = re.compile(
+    r"^[ \t]*(?:template\s*<[^>]*>\s*)?"
+    r"(?:[A-Za-z_][\w:<>,\s\*&]*?[\s\*&])"
+    r"([A-Za-z_]\w*)\s*\([^;{)]*\)\s*(?:const\s*)?\{", re.M)
+
 RE_LOCAL_MACRO = re.compile(
     r"^[ \t]*#[ \t]*define[ \t]+([A-Za-z_]\w*)[ \t]*\(",
     re.M)
@@ -275,5 +288,45 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
             is_parametrized=parametrized,
             param_multiplicity=max(inst_counts.get(suite, 0), 1)
             if parametrized else 1,
+        ))
+    if not cases:
+        cases = _extract_freestanding(text, rel)
+    return cases
+
+
+def _extract_freestanding(text: str, rel: str) -> List[TestCase]:
+    """Fallback for hand-rolled C++ test files with no framework macro:
+    one case per top-level function containing CHECK-family assertions
+    (glog/openfst style; per-function granularity matches how the study
+    labeled those suites as components like 'WeightTester')."""
+    cases: List[TestCase] = []
+    for m in RE_FUNC_DEF.finditer(text):
+        brace = text.rfind("{", m.start(), m.end())
+        end = _match_brace_block(text, brace)
+        body = text[m.start():end]
+        lineno = text.count("\n", 0, m.start()) + 1
+        assertions: List[Assertion] = []
+        for am in RE_CHECK.finditer(body):
+            call = am.group(1)
+            src = call + _extract_call(body, am.end() - 1)
+            a_line = lineno + body.count("\n", 0, am.start())
+            kind = "approx" if ("NEAR" in call or "CLOSE" in call) \
+                else "unittest"
+            assertions.append(Assertion(
+                kind=kind, call_name=call, source=src[:500], lineno=a_line,
+                exception=""))
+        if not assertions:
+            continue
+        cases.append(TestCase(
+            name=m.group(1),
+            qualname=m.group(1),
+            file_rel=rel,
+            lineno=lineno,
+            end_lineno=lineno + body.count("\n"),
+            source=body[:4000],
+            assertions=assertions,
+            uses_mock=False,
+            is_parametrized=False,
+            param_multiplicity=1,
         ))
     return cases
