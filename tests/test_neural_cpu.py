@@ -35,3 +35,35 @@ def test_split_seed_pins_partition():
     _, va1 = ds.split(val_frac=0.2, seed=7)
     _, va2 = ds.split(val_frac=0.2, seed=7)
     assert va1.texts == va2.texts
+
+
+def test_ensemble_probs_script(tmp_path):
+    """scripts/ensemble_probs.py: shared-gold verification + averaging."""
+    import subprocess
+    import sys
+
+    import torch
+    g = torch.Generator().manual_seed(0)
+    gold = {"strategy": (torch.rand(20, 19, generator=g) < 0.2).float(),
+            "property": (torch.rand(20, 21, generator=g) < 0.2).float()}
+    paths = []
+    for s in range(2):
+        probs = {h: torch.rand(*gold[h].shape, generator=g) * 0.5 +
+                 gold[h] * 0.4 for h in gold}
+        p = tmp_path / f"p{s}.pt"
+        torch.save({"val_probs": probs, "val_gold": gold, "seed": s}, p)
+        paths.append(str(p))
+    out = subprocess.run(
+        [sys.executable, "scripts/ensemble_probs.py"] + paths,
+        capture_output=True, text=True, cwd="/root/repo")
+    assert out.returncode == 0, out.stderr
+    assert "ENSEMBLE x2" in out.stdout
+    # mismatched gold must abort
+    bad = {"val_probs": {h: torch.rand(*gold[h].shape) for h in gold},
+           "val_gold": {h: 1 - gold[h] for h in gold}, "seed": 9}
+    pb = tmp_path / "bad.pt"
+    torch.save(bad, pb)
+    out2 = subprocess.run(
+        [sys.executable, "scripts/ensemble_probs.py", paths[0], str(pb)],
+        capture_output=True, text=True, cwd="/root/repo")
+    assert out2.returncode != 0
