@@ -86,3 +86,83 @@ def test_tokenizer_ids_in_vocab(src):
     ids = tok.encode(src, 64)
     assert len(ids) <= 64
     assert all(0 <= i < 1024 for i in ids)
+
+
+def _mini_xlsx(tmp, cells_xml, shared_xml=None):
+    import zipfile
+    p = str(tmp / "f.xlsx")
+    with zipfile.ZipFile(p, "w") as z:
+        z.writestr("[Content_Types].xml", "<Types/>")
+        z.writestr(
+            "xl/workbook.xml",
+            '<workbook xmlns="http://schemas.openxmlformats.org/'
+            'spreadsheetml/2006/main" xmlns:r="http://schemas.'
+            'openxmlformats.org/officeDocument/2006/relationships">'
+            '<sheets><sheet name="S" sheetId="1" r:id="rId1"/></sheets>'
+            '</workbook>')
+        z.writestr(
+            "xl/_rels/workbook.xml.rels",
+            '<Relationships xmlns="http://schemas.openxmlformats.org/'
+            'package/2006/relationships"><Relationship Id="rId1" '
+            'Type="x" Target="worksheets/sheet1.xml"/></Relationships>')
+        if shared_xml is not None:
+            z.writestr("xl/sharedStrings.xml", shared_xml)
+        z.writestr(
+            "xl/worksheets/sheet1.xml",
+            '<worksheet xmlns="http://schemas.openxmlformats.org/'
+            'spreadsheetml/2006/main"><sheetData>%s</sheetData>'
+            '</worksheet>' % cells_xml)
+    return p
+
+
+# XML 1.0 cannot carry most control chars even escaped, and \r
+# normalizes to \n per spec — keep the alphabet XML-transparent
+_XML_SAFE = (string.ascii_letters + string.digits +
+             " \t\n!#$%&'()*+,-./:;<=>?@[]^_`{|}~\"")
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.text(alphabet=_XML_SAFE, max_size=60),
+       st.integers(min_value=0, max_value=40))
+def test_xlsx_reader_fuzz(tmp_path_factory, text, col):
+    """read_xlsx is total over arbitrary cell text and sparse columns,
+    and round-trips inline-string content."""
+    from xml.sax.saxutils import escape
+
+    from tosem2021_amd.utils.xlsx import read_xlsx
+    tmp = tmp_path_factory.mktemp("xl")
+    letters = ""
+    c = col
+    while True:
+        letters = chr(ord("A") + c % 26) + letters
+        c = c // 26 - 1
+        if c < 0:
+            break
+    cells = ('<row r="1"><c r="%s1" t="inlineStr"><is><t>%s</t></is></c>'
+             '</row>' % (letters, escape(text)))
+    rows = read_xlsx(_mini_xlsx(tmp, cells))["S"]
+    assert len(rows) == 1 and len(rows[0]) == col + 1
+    assert rows[0][col] == text
+    for k in range(col):
+        assert rows[0][k] == ""
+
+
+def test_xlsx_reader_hostile_inputs(tmp_path):
+    """Malformed parts degrade, never crash: truncated zip, bad XML,
+    out-of-range shared-string index."""
+    import zipfile
+
+    import pytest as _pytest
+
+    from tosem2021_amd.utils.xlsx import read_xlsx
+    bad = tmp_path / "bad.xlsx"
+    bad.write_bytes(b"PK\x03\x04 not a zip really")
+    with _pytest.raises((zipfile.BadZipFile, KeyError, OSError)):
+        read_xlsx(str(bad))
+    # shared-string index beyond the table
+    p = _mini_xlsx(tmp_path,
+                   '<row r="1"><c r="A1" t="s"><v>99</v></c></row>',
+                   '<sst xmlns="http://schemas.openxmlformats.org/'
+                   'spreadsheetml/2006/main"><si><t>only</t></si></sst>')
+    rows = read_xlsx(p)["S"]
+    assert len(rows) == 1   # degrades to empty/raw, no IndexError
