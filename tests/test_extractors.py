@@ -311,3 +311,58 @@ def test_walker_admits_tester_headers():
     from tosem2021_amd.corpus.walker import SKIP_DIRS
     assert "third_party" not in SKIP_DIRS
     assert "node_modules" in SKIP_DIRS
+
+
+def test_python_assertion_helper_expansion(tmp_path):
+    """Assertions inside file-local helper methods count at their call
+    sites in test cases (DeepSpeech test_value_range.py pattern)."""
+    src = '''
+import unittest
+
+class T(unittest.TestCase):
+    def _ending_tester(self, v, e):
+        r = compute(v)
+        self.assertEqual(r, e)
+
+    def test_scalar(self):
+        self._ending_tester(1, 1)
+        self._ending_tester(2, 2)
+
+    def test_direct(self):
+        self.assertTrue(ok())
+'''
+    p = tmp_path / "test_vr.py"
+    p.write_text(src)
+    from tosem2021_amd.extract.python_extractor import extract_file
+    cases = {c.name: c for c in extract_file(str(p))}
+    assert len(cases["test_scalar"].assertions) == 2
+    assert cases["test_scalar"].assertions[0].call_name == "assertEqual"
+    assert len(cases["test_direct"].assertions) == 1
+
+
+def test_gtest_helper_function_expansion(tmp_path):
+    """C++ helper functions holding the assertions (kenlm model_test.cc
+    pattern, incl. templated calls and helper->helper nesting) are
+    expanded transitively into the calling case."""
+    src = """
+template <class M> void Starters(const M &m) {
+  BOOST_CHECK_EQUAL(m.Size(), 1);
+  BOOST_CHECK(m.Ok());
+}
+template <class M> void Everything(const M &m) {
+  Starters<M>(m);
+}
+BOOST_AUTO_TEST_CASE(probing) {
+  ProbingModel m("f.arpa");
+  Everything<ProbingModel>(m);
+}
+BOOST_AUTO_TEST_CASE(direct) {
+  BOOST_REQUIRE(true);
+}
+"""
+    p = tmp_path / "model_test.cc"
+    p.write_text(src)
+    from tosem2021_amd.extract.gtest_extractor import extract_gtest_file
+    cases = {c.name: c for c in extract_gtest_file(str(p))}
+    assert len(cases["probing"].assertions) == 2
+    assert len(cases["direct"].assertions) == 1

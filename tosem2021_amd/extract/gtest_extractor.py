@@ -235,6 +235,53 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
         marks.append((m, m.group(2), m.group(3)))
     for m in RE_BOOST_CASE.finditer(text):
         marks.append((m, "", m.group(2)))
+    # one-level assertion-helper expansion (same idea as local macros):
+    # a file-local function whose body asserts (kenlm model_test.cc
+    # StartTest -> BOOST_CHECK/SLOPPY_CHECK_CLOSE) makes its call sites
+    # inside a case count as that helper's assertions.
+    helper_asserts = {}
+    helper_bodies = {}
+    case_names = {name for _, _, name in marks}
+    for fm in RE_FUNC_DEF.finditer(text):
+        fname = fm.group(1)
+        if fname in case_names or fname in ("main", "if", "for", "while",
+                                            "switch"):
+            continue
+        fb = text.rfind("{", fm.start(), fm.end())
+        fbody = text[fm.start():_match_brace_block(text, fb)]
+        found = [(am.group(1), am.end() - 1) for am in
+                 RE_ASSERT.finditer(fbody)]
+        if re_local is not None:
+            found += [(local_asserts[am.group(1)], am.end() - 1) for am in
+                      re_local.finditer(fbody)]
+        helper_asserts[fname] = [
+            (call, (call + _extract_call(fbody, pos))[:500])
+            for call, pos in found]
+        helper_bodies[fname] = fbody
+    # propagate through helper->helper calls (kenlm: case -> Everything ->
+    # Starters -> BOOST_CHECK), bounded depth, no self-recursion blowup
+    for _ in range(3):
+        changed = False
+        for fname, fbody in helper_bodies.items():
+            inherited = []
+            for other, asserts in helper_asserts.items():
+                if other != fname and asserts and \
+                        re.search(r"\b" + re.escape(other) +
+                                  r"\s*(?:<[^;{}()]*>)?\s*\(", fbody):
+                    inherited.extend(asserts)
+            merged = helper_asserts.get(fname, []) + [
+                a for a in inherited
+                if a not in helper_asserts.get(fname, [])]
+            if len(merged) > len(helper_asserts.get(fname, [])) and \
+                    len(merged) <= 200:
+                helper_asserts[fname] = merged
+                changed = True
+        if not changed:
+            break
+    helper_asserts = {k: v for k, v in helper_asserts.items() if v}
+    re_helper = (re.compile(
+        r"\b(" + "|".join(map(re.escape, sorted(helper_asserts))) +
+        r")\s*(?:<[^;{}()]*>)?\s*\(") if helper_asserts else None)
     for m, suite, name in sorted(marks, key=lambda t: t[0].start()):
         brace = text.find("{", m.end())
         if brace < 0:
@@ -272,6 +319,21 @@ def extract_gtest_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
                 assertions.append(Assertion(
                     kind=kind, call_name=call, source=src[:500],
                     lineno=a_line, exception=""))
+            assertions.sort(key=lambda a: a.lineno)
+        if re_helper is not None:
+            body_off = text.find("{", m.end())
+            inner = text[body_off:end] if body_off >= 0 else ""
+            for hm in re_helper.finditer(inner):
+                a_line = lineno + body[:len(body)].count(
+                    "\n", 0, body_off - m.start() + hm.start())
+                for call, src in helper_asserts[hm.group(1)]:
+                    kind = "raises" if RE_DEATH.search(call) else "unittest"
+                    if "NEAR" in call or "FLOAT_EQ" in call \
+                            or "DOUBLE_EQ" in call or "CLOSE" in call:
+                        kind = "approx"
+                    assertions.append(Assertion(
+                        kind=kind, call_name=call, source=src,
+                        lineno=a_line, exception=""))
             assertions.sort(key=lambda a: a.lineno)
         macro = m.group(1) if m.re is RE_TEST_MACRO else ""
         parametrized = macro in ("TEST_P", "TYPED_TEST_P")

@@ -147,6 +147,42 @@ def extract_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
     lines = text.splitlines()
     cases: List[TestCase] = []
 
+    # one-level assertion-helper expansion (mirrors the gtest extractor's
+    # local-macro handling): a non-test function/method in this file whose
+    # body asserts (e.g. DeepSpeech tests/test_value_range.py
+    # `_ending_tester` -> assertEqual) makes each of its call sites inside
+    # a test case count as that helper's assertions.
+    helper_asserts = {}
+
+    def collect_helpers(body):
+        for node in body:
+            if isinstance(node, ast.ClassDef):
+                collect_helpers(node.body)
+            elif isinstance(node, (ast.FunctionDef, ast.AsyncFunctionDef)):
+                if _is_test_func(node.name, True):
+                    continue
+                hv = _AssertVisitor(lines)
+                hv.visit(node)
+                if hv.assertions:
+                    helper_asserts[node.name] = hv.assertions
+
+    collect_helpers(tree.body)
+
+    def helper_calls(node):
+        out = []
+        for sub in ast.walk(node):
+            if isinstance(sub, ast.Call):
+                fn = sub.func
+                name = fn.attr if isinstance(fn, ast.Attribute) else \
+                    (fn.id if isinstance(fn, ast.Name) else "")
+                if name in helper_asserts:
+                    for a in helper_asserts[name]:
+                        out.append(Assertion(
+                            kind=a.kind, call_name=a.call_name,
+                            source=a.source, lineno=sub.lineno,
+                            exception=a.exception))
+        return out
+
     def visit_body(body, class_name: str, in_test_class: bool):
         for node in body:
             if isinstance(node, ast.ClassDef):
@@ -158,6 +194,8 @@ def extract_file(path: str, rel: Optional[str] = None) -> List[TestCase]:
                     continue
                 v = _AssertVisitor(lines)
                 v.visit(node)
+                v.assertions.extend(helper_calls(node))
+                v.assertions.sort(key=lambda a: a.lineno)
                 decs = _decorator_names(node)
                 seg = ast.get_source_segment(text, node) or ""
                 cases.append(TestCase(
