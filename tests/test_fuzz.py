@@ -166,3 +166,21 @@ def test_xlsx_reader_hostile_inputs(tmp_path):
                    'spreadsheetml/2006/main"><si><t>only</t></si></sst>')
     rows = read_xlsx(p)["S"]
     assert len(rows) == 1   # degrades to empty/raw, no IndexError
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.tuples(
+    st.sampled_from(["Apollo", "Ray", "nni", "tpot"]),
+    st.floats(min_value=0.01, max_value=99.99)), min_size=0, max_size=4,
+    unique_by=lambda t: t[0]))
+def test_mirror_cell_encoding_roundtrip(pairs):
+    """The `repo:(x%)` correlation-cell encoding written by the mirror is
+    parsed back losslessly by golden_mirror's encoded-cell extractor."""
+    from tosem2021_amd.analyze.golden_mirror import _encoded_cells
+    cell = ",".join(f"{r}:({v:.2f}%)" for r, v in pairs)
+    rows = [["strategy", "col"], ["negative_test", cell]]
+    out = _encoded_cells(rows)
+    assert len(out) == len(pairs)
+    for r, v in pairs:
+        got = out[("negative_test", 0, r)]
+        assert abs(round(v, 2) - got) < 0.005
